@@ -1,0 +1,111 @@
+"""Distributed plumbing tier: gloo, world_size=2, CPU.
+
+Covers the C1-C7 call sites, cross-rank theta bitwise identity, ws=1 vs ws=2
+numerical equivalence, and resume-from-checkpoint under a world.
+"""
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from distributedes_amd import preset
+
+
+def _cfg(**kw):
+    base = dict(pop=8, generations=3, episode_steps=25,
+                noise_table_size=1 << 16, device="cpu", backend="gloo",
+                obs_normalize=True)
+    base.update(kw)
+    return preset("cartpole_es", **base)
+
+
+def _worker(rank, world, port, fn_name, out_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    try:
+        result = globals()[fn_name](rank, world)
+        out_q.put((rank, "ok", result))
+    except Exception as e:  # pragma: no cover
+        import traceback
+        out_q.put((rank, "err", traceback.format_exc()))
+    finally:
+        import torch.distributed as dist
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def _spawn(fn_name, world=2, port=29600):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_worker, args=(r, world, port, fn_name, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, status, payload = q.get()
+        assert status == "ok", payload
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=60)
+    return results
+
+
+# --- worker bodies ---------------------------------------------------------
+
+def _body_train3(rank, world):
+    from distributedes_amd.core.engine import TrainState, train_step
+    st = TrainState(_cfg(), ops_backend="torch_ref")
+    rets = []
+    for _ in range(3):
+        rets.append(train_step(st)["episodic_return_mean"])
+    return {"theta": st.theta.clone(), "rets": rets,
+            "norm_count": float(st.normalizer.count)}
+
+
+def _body_resume(rank, world):
+    import tempfile
+    from distributedes_amd.core.engine import TrainState, train_step
+    from distributedes_amd.utils import load_checkpoint, save_checkpoint
+    st = TrainState(_cfg(), ops_backend="torch_ref")
+    for _ in range(2):
+        train_step(st)
+    # each rank snapshots (identical) state, reloads into a fresh world member
+    payload = st.checkpoint_payload()
+    for _ in range(1):
+        train_step(st)
+    st2 = TrainState(_cfg(), ops_backend="torch_ref")
+    st2.load_payload(payload)
+    train_step(st2)
+    return {"same": torch.equal(st.theta, st2.theta)}
+
+
+def test_ws2_ranks_bitwise_identical_and_match_ws1():
+    res = _spawn("_body_train3", world=2, port=29611)
+    # ranks agree bitwise (C4 all-reduce produces the same bits everywhere)
+    assert torch.equal(res[0]["theta"], res[1]["theta"])
+    assert res[0]["rets"] == res[1]["rets"]
+    # and the 2-rank run numerically matches a single-process run
+    from distributedes_amd.core.engine import TrainState, train_step
+    for k in ("WORLD_SIZE", "RANK"):
+        os.environ.pop(k, None)
+    st = TrainState(_cfg(backend=None), ops_backend="torch_ref")
+    rets = []
+    for _ in range(3):
+        rets.append(train_step(st)["episodic_return_mean"])
+    torch.testing.assert_close(st.theta, res[0]["theta"], rtol=1e-4, atol=1e-5)
+    # generation 1 precedes any cross-rank float reassociation: exact match.
+    # (later generations may flip a discrete argmax action from low-order
+    # reduction-order bits, so theta closeness is the right check there)
+    assert rets[0] == res[0]["rets"][0]
+    # normalizer saw the same global observation count
+    assert float(st.normalizer.count) == res[0]["norm_count"]
+
+
+def test_ws2_resume_bit_equivalence():
+    res = _spawn("_body_resume", world=2, port=29627)
+    assert res[0]["same"] and res[1]["same"]
