@@ -72,9 +72,11 @@ def evaluate(checkpoint_or_state, config: Optional[Config] = None,
         state.load_payload(load_checkpoint(checkpoint_or_state))
 
     P = episodes
-    pop_w = state.theta.unsqueeze(0).expand(P, -1).contiguous()
-    if state.device.type == "cuda" and state.cfg.precision == "bf16":
-        pop_w = pop_w.to(torch.bfloat16)
+    dt = (torch.bfloat16 if state.device.type == "cuda"
+          and state.cfg.precision == "bf16" else torch.float32)
+    stride = ((state.dim + 7) & ~7) if state._hip_rollout is not None else state.dim
+    pop_w = torch.zeros(P, stride, dtype=dt, device=state.device)
+    pop_w[:, :state.dim] = state.theta.to(dt)
     # distinct env streams per episode
     saved_base, saved_pop = state.member_base, state.local_pop
     state.member_base, state.local_pop = 0, P

@@ -108,6 +108,14 @@ class TrainState:
                                      log_every=cfg.log_every)
         self._pop_w = None          # reusable [local_pop, D] perturbed weights
 
+        # GPU-native rollout path (K3/K5/K7/K8 kernels); torch loop otherwise
+        self._hip_rollout = None
+        if self.ops.name == "hip":
+            from ..ops.rollout import HipRollout
+            if HipRollout.supports(cfg, self.model):
+                self._hip_rollout = HipRollout(cfg, self.env, self.model,
+                                               self.ops.ext, self.device)
+
     # ------------------------------------------------------------------ #
     def member_offsets_signs(self, generation: int):
         """Offsets/signs for the LOCAL member slice, plus global arrays."""
@@ -135,6 +143,11 @@ class TrainState:
         cfg = self.cfg
         P = pop_w.shape[0]
         gen_seed = noise_mod.splitmix64(cfg.seed ^ (self.generation * 0x9E37)) & 0x7FFFFFFF
+        if self._hip_rollout is not None:
+            return self._hip_rollout.run(pop_w, gen_seed, self.env_ids(),
+                                         cfg.episode_steps, self.normalizer)
+        if pop_w.shape[1] != self.dim:           # padded rows -> logical view
+            pop_w = pop_w[:, :self.dim]
         obs = self.env.reset(P, gen_seed, device=self.device, env_ids=self.env_ids())
         fitness = torch.zeros(P, dtype=torch.float32, device=self.device)
         alive = torch.ones(P, dtype=torch.bool, device=self.device)
@@ -165,9 +178,12 @@ class TrainState:
                      and self.device.type == "cuda") else torch.float32)
         base = self.theta
         sigma_vec = self.sigma_vec if cfg.algo == "cem" else None
+        kw = {}
+        if self.ops.name == "hip":
+            kw["pad_rows"] = self._hip_rollout is not None
         self._pop_w = self.ops.perturb(base, self.table, offsets, signs,
                                        cfg.sigma, sigma_vec=sigma_vec,
-                                       out_dtype=out_dtype)
+                                       out_dtype=out_dtype, **kw)
         return self._pop_w
 
     # ------------------------------------------------------------------ #

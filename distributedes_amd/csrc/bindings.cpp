@@ -6,18 +6,51 @@
 #include <torch/extension.h>
 
 #include <cstdint>
+#include <vector>
 
 extern "C" {
 void des_launch_noise_fill(float*, int64_t, uint64_t, hipStream_t);
 void des_launch_perturb_f32(float*, const float*, const float*, const int64_t*,
                             const float*, float, const float*, int64_t,
-                            int64_t, hipStream_t);
+                            int64_t, int64_t, hipStream_t);
 void des_launch_perturb_bf16(uint16_t*, const float*, const float*,
                              const int64_t*, const float*, float, const float*,
-                             int64_t, int64_t, hipStream_t);
+                             int64_t, int64_t, int64_t, hipStream_t);
 void des_launch_wsum(float*, float*, const float*, const int64_t*,
                      const float*, const float*, int64_t, int64_t, bool,
                      hipStream_t);
+void des_launch_mlp_forward(const void*, bool, int64_t, const float*, int,
+                            const float*, const float*, float, int*, float*,
+                            float*, int, int, int, const int64_t*, int,
+                            hipStream_t);
+void des_launch_nature_cnn_forward(const uint16_t*, int64_t, const void*, bool,
+                                   const int*, int*, float*, int, int,
+                                   const int64_t*, hipStream_t);
+void des_launch_cartpole_reset(float*, int*, int64_t*, float*, uint8_t*,
+                               float*, const int64_t*, int64_t, int,
+                               hipStream_t);
+void des_launch_cartpole_step(float*, int*, int64_t*, const int*, float*,
+                              uint8_t*, float*, const int64_t*, int64_t, int,
+                              hipStream_t);
+void des_launch_pong_reset(float*, float*, float*, float*, float*, float*,
+                           int*, int*, int64_t*, uint8_t*, float*, uint8_t*,
+                           float*, const int64_t*, int64_t, int, hipStream_t);
+void des_launch_pong_step(float*, float*, float*, float*, float*, float*,
+                          int*, int*, int64_t*, uint8_t*, int, const int*,
+                          float*, uint8_t*, float*, const int64_t*, int64_t,
+                          int, hipStream_t);
+void des_launch_latent_reset(float*, int*, float*, uint8_t*, float*,
+                             const int64_t*, int64_t, int, hipStream_t);
+void des_launch_latent_render(const float*, uint16_t*, int, int, int,
+                              hipStream_t);
+void des_launch_latent_step(float*, int*, const int*, float*, uint8_t*,
+                            float*, int, int, int, hipStream_t);
+void des_launch_humanoid_reset(float*, int*, float*, uint8_t*, float*,
+                               const int64_t*, int64_t, int, hipStream_t);
+void des_launch_humanoid_step(float*, int*, const float*, float*, uint8_t*,
+                              float*, int, int, hipStream_t);
+void des_launch_normalizer_observe(const float*, double*, double*, int, int,
+                                   hipStream_t);
 }
 
 namespace {
@@ -28,6 +61,12 @@ inline hipStream_t cur_stream() {
 
 #define CHECK_DEV(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
 #define CHECK_CONT(t) TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+float* fptr(torch::Tensor& t) { return t.data_ptr<float>(); }
+const float* cfptr(const torch::Tensor& t) { return t.data_ptr<float>(); }
+int* iptr(torch::Tensor& t) { return t.data_ptr<int>(); }
+int64_t* lptr(torch::Tensor& t) { return t.data_ptr<int64_t>(); }
+uint8_t* bptr(torch::Tensor& t) { return t.data_ptr<uint8_t>(); }
 
 void noise_fill(torch::Tensor table, int64_t seed) {
   CHECK_DEV(table); CHECK_CONT(table);
@@ -42,8 +81,8 @@ void perturb(torch::Tensor out, torch::Tensor theta, torch::Tensor table,
   CHECK_DEV(out); CHECK_CONT(out); CHECK_DEV(theta); CHECK_CONT(theta);
   CHECK_DEV(table); CHECK_CONT(table);
   CHECK_DEV(offsets); CHECK_CONT(offsets); CHECK_DEV(signs); CHECK_CONT(signs);
-  int64_t P = out.size(0), D = out.size(1);
-  TORCH_CHECK(theta.numel() == D && offsets.numel() == P && signs.numel() == P);
+  int64_t P = out.size(0), stride = out.size(1), D = theta.numel();
+  TORCH_CHECK(stride >= D && offsets.numel() == P && signs.numel() == P);
   TORCH_CHECK(offsets.dtype() == torch::kInt64 && signs.dtype() == torch::kFloat32);
   const float* sv = nullptr;
   if (sigma_vec.numel() > 0) {
@@ -55,12 +94,12 @@ void perturb(torch::Tensor out, torch::Tensor theta, torch::Tensor table,
     des_launch_perturb_bf16(
         reinterpret_cast<uint16_t*>(out.data_ptr()), theta.data_ptr<float>(),
         table.data_ptr<float>(), offsets.data_ptr<int64_t>(),
-        signs.data_ptr<float>(), (float)sigma, sv, P, D, cur_stream());
+        signs.data_ptr<float>(), (float)sigma, sv, P, D, stride, cur_stream());
   } else if (out.dtype() == torch::kFloat32) {
     des_launch_perturb_f32(out.data_ptr<float>(), theta.data_ptr<float>(),
                            table.data_ptr<float>(),
                            offsets.data_ptr<int64_t>(), signs.data_ptr<float>(),
-                           (float)sigma, sv, P, D, cur_stream());
+                           (float)sigma, sv, P, D, stride, cur_stream());
   } else {
     TORCH_CHECK(false, "perturb: out must be bf16 or fp32");
   }
@@ -81,6 +120,160 @@ void weighted_noise_sum(torch::Tensor acc, torch::Tensor acc_sq,
                   want_sq, cur_stream());
 }
 
+// ---------------- policy forwards ----------------
+void mlp_forward(torch::Tensor popw, torch::Tensor obs, torch::Tensor nmean,
+                 torch::Tensor ninv, double clip, torch::Tensor actions,
+                 torch::Tensor logits, int64_t h1, int64_t h2, int64_t A,
+                 std::vector<int64_t> offs) {
+  CHECK_DEV(popw); CHECK_CONT(popw); CHECK_DEV(obs); CHECK_CONT(obs);
+  CHECK_DEV(actions); CHECK_CONT(actions);
+  TORCH_CHECK(offs.size() == 6);
+  int64_t P = popw.size(0);
+  int in_dim = obs.size(1);
+  TORCH_CHECK(obs.size(0) == P && obs.dtype() == torch::kFloat32);
+  const float* nm = nullptr;
+  const float* ni = nullptr;
+  if (nmean.numel() > 0) {
+    TORCH_CHECK(nmean.numel() == in_dim && ninv.numel() == in_dim);
+    nm = cfptr(nmean); ni = cfptr(ninv);
+  }
+  bool bf16 = popw.dtype() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || popw.dtype() == torch::kFloat32);
+  int* ai = nullptr;
+  float* af = nullptr;
+  if (actions.dtype() == torch::kInt32) {
+    TORCH_CHECK(actions.numel() == P);
+    ai = iptr(actions);
+  } else {
+    TORCH_CHECK(actions.dtype() == torch::kFloat32 && actions.numel() == P * A);
+    af = fptr(actions);
+  }
+  float* lo = logits.numel() > 0 ? fptr(logits) : nullptr;
+  des_launch_mlp_forward(popw.data_ptr(), bf16, popw.size(1), cfptr(obs),
+                         in_dim, nm, ni, (float)clip, ai, af, lo, (int)h1,
+                         (int)h2, (int)A, offs.data(), (int)P, cur_stream());
+}
+
+void nature_cnn_forward(torch::Tensor popw, torch::Tensor frames,
+                        torch::Tensor slot_map, torch::Tensor actions,
+                        torch::Tensor logits, int64_t A, torch::Tensor offs) {
+  CHECK_DEV(popw); CHECK_CONT(popw); CHECK_DEV(frames); CHECK_CONT(frames);
+  CHECK_DEV(slot_map); CHECK_DEV(actions); CHECK_DEV(offs);
+  TORCH_CHECK(popw.dtype() == torch::kBFloat16, "pop weights must be bf16");
+  TORCH_CHECK(offs.dtype() == torch::kInt64 && offs.numel() == 10);
+  TORCH_CHECK(slot_map.dtype() == torch::kInt32 && slot_map.numel() == 4);
+  TORCH_CHECK(actions.dtype() == torch::kInt32);
+  int64_t P = popw.size(0);
+  bool u8 = frames.dtype() == torch::kUInt8;
+  TORCH_CHECK(u8 || frames.dtype() == torch::kBFloat16);
+  float* lo = logits.numel() > 0 ? fptr(logits) : nullptr;
+  des_launch_nature_cnn_forward(
+      reinterpret_cast<const uint16_t*>(popw.data_ptr()), popw.size(1),
+      frames.data_ptr(), u8, slot_map.data_ptr<int>(), iptr(actions), lo,
+      (int)A, (int)P, offs.data_ptr<int64_t>(), cur_stream());
+}
+
+// ---------------- envs ----------------
+void cartpole_reset(torch::Tensor state, torch::Tensor steps,
+                    torch::Tensor reset_ctr, torch::Tensor fitness,
+                    torch::Tensor alive, torch::Tensor ep_len,
+                    torch::Tensor env_ids, int64_t seed_mix) {
+  int P = state.size(0);
+  des_launch_cartpole_reset(fptr(state), iptr(steps), lptr(reset_ctr),
+                            fptr(fitness), bptr(alive), fptr(ep_len),
+                            lptr(env_ids), seed_mix, P, cur_stream());
+}
+
+void cartpole_step(torch::Tensor state, torch::Tensor steps,
+                   torch::Tensor reset_ctr, torch::Tensor actions,
+                   torch::Tensor fitness, torch::Tensor alive,
+                   torch::Tensor ep_len, torch::Tensor env_ids,
+                   int64_t seed_mix) {
+  int P = state.size(0);
+  des_launch_cartpole_step(fptr(state), iptr(steps), lptr(reset_ctr),
+                           iptr(actions), fptr(fitness), bptr(alive),
+                           fptr(ep_len), lptr(env_ids), seed_mix, P,
+                           cur_stream());
+}
+
+void pong_reset(torch::Tensor bx, torch::Tensor by, torch::Tensor vx,
+                torch::Tensor vy, torch::Tensor py, torch::Tensor oy,
+                torch::Tensor score, torch::Tensor points,
+                torch::Tensor serve_ctr, torch::Tensor frames,
+                torch::Tensor fitness, torch::Tensor alive,
+                torch::Tensor ep_len, torch::Tensor env_ids, int64_t seed) {
+  int P = bx.size(0);
+  des_launch_pong_reset(fptr(bx), fptr(by), fptr(vx), fptr(vy), fptr(py),
+                        fptr(oy), iptr(score), iptr(points), lptr(serve_ctr),
+                        bptr(frames), fptr(fitness), bptr(alive), fptr(ep_len),
+                        lptr(env_ids), seed, P, cur_stream());
+}
+
+void pong_step(torch::Tensor bx, torch::Tensor by, torch::Tensor vx,
+               torch::Tensor vy, torch::Tensor py, torch::Tensor oy,
+               torch::Tensor score, torch::Tensor points,
+               torch::Tensor serve_ctr, torch::Tensor frames, int64_t slot,
+               torch::Tensor actions, torch::Tensor fitness,
+               torch::Tensor alive, torch::Tensor ep_len,
+               torch::Tensor env_ids, int64_t seed) {
+  int P = bx.size(0);
+  des_launch_pong_step(fptr(bx), fptr(by), fptr(vx), fptr(vy), fptr(py),
+                       fptr(oy), iptr(score), iptr(points), lptr(serve_ctr),
+                       bptr(frames), (int)slot, iptr(actions), fptr(fitness),
+                       bptr(alive), fptr(ep_len), lptr(env_ids), seed, P,
+                       cur_stream());
+}
+
+void latent_reset(torch::Tensor z, torch::Tensor steps, torch::Tensor fitness,
+                  torch::Tensor alive, torch::Tensor ep_len,
+                  torch::Tensor env_ids, int64_t seed) {
+  des_launch_latent_reset(fptr(z), iptr(steps), fptr(fitness), bptr(alive),
+                          fptr(ep_len), lptr(env_ids), seed, z.size(0),
+                          cur_stream());
+}
+
+void latent_render(torch::Tensor z, torch::Tensor obs) {
+  TORCH_CHECK(obs.dtype() == torch::kBFloat16);
+  int C = obs.size(1), HW = obs.size(2);
+  des_launch_latent_render(cfptr(z),
+                           reinterpret_cast<uint16_t*>(obs.data_ptr()), C, HW,
+                           z.size(0), cur_stream());
+}
+
+void latent_step(torch::Tensor z, torch::Tensor steps, torch::Tensor actions,
+                 torch::Tensor fitness, torch::Tensor alive,
+                 torch::Tensor ep_len, int64_t num_actions, int64_t horizon) {
+  des_launch_latent_step(fptr(z), iptr(steps), iptr(actions), fptr(fitness),
+                         bptr(alive), fptr(ep_len), (int)num_actions,
+                         (int)horizon, z.size(0), cur_stream());
+}
+
+void humanoid_reset(torch::Tensor s, torch::Tensor steps,
+                    torch::Tensor fitness, torch::Tensor alive,
+                    torch::Tensor ep_len, torch::Tensor env_ids,
+                    int64_t seed) {
+  des_launch_humanoid_reset(fptr(s), iptr(steps), fptr(fitness), bptr(alive),
+                            fptr(ep_len), lptr(env_ids), seed, s.size(0),
+                            cur_stream());
+}
+
+void humanoid_step(torch::Tensor s, torch::Tensor steps, torch::Tensor actions,
+                   torch::Tensor fitness, torch::Tensor alive,
+                   torch::Tensor ep_len, int64_t horizon) {
+  des_launch_humanoid_step(fptr(s), iptr(steps), cfptr(actions), fptr(fitness),
+                           bptr(alive), fptr(ep_len), (int)horizon, s.size(0),
+                           cur_stream());
+}
+
+void normalizer_observe(torch::Tensor obs, torch::Tensor sum,
+                        torch::Tensor sumsq) {
+  CHECK_DEV(obs); CHECK_CONT(obs);
+  TORCH_CHECK(sum.dtype() == torch::kFloat64 && sumsq.dtype() == torch::kFloat64);
+  des_launch_normalizer_observe(cfptr(obs), sum.data_ptr<double>(),
+                                sumsq.data_ptr<double>(), obs.size(0),
+                                obs.size(1), cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -88,4 +281,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("perturb", &perturb, "K2: theta + sigma*sign*noise-slice");
   m.def("weighted_noise_sum", &weighted_noise_sum,
         "K10: sum_i w_i s_i eps_i (+ optional sum w_i eps_i^2)");
+  m.def("mlp_forward", &mlp_forward, "K3: per-member MLP forward + action");
+  m.def("nature_cnn_forward", &nature_cnn_forward,
+        "K5: per-member Nature-CNN forward + argmax");
+  m.def("cartpole_reset", &cartpole_reset);
+  m.def("cartpole_step", &cartpole_step);
+  m.def("pong_reset", &pong_reset);
+  m.def("pong_step", &pong_step);
+  m.def("latent_reset", &latent_reset);
+  m.def("latent_render", &latent_render);
+  m.def("latent_step", &latent_step);
+  m.def("humanoid_reset", &humanoid_reset);
+  m.def("humanoid_step", &humanoid_step);
+  m.def("normalizer_observe", &normalizer_observe);
 }

@@ -91,6 +91,13 @@ __device__ __forceinline__ float hash_uniform(int64_t counter, float lo, float h
   return lo + (hi - lo) * u;
 }
 
+// full-wave sum reduction; every lane returns the sum
+__device__ __forceinline__ float wave_reduce_sum_f(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off);
+  return __shfl(v, 0);
+}
+
 // ---------------- bf16 helpers ----------------
 __device__ __forceinline__ uint16_t f32_to_bf16(float f) {
   uint32_t b = __float_as_uint(f);

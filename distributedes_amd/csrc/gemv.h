@@ -1,0 +1,58 @@
+// Wave-cooperative GEMV helpers for per-member policy forwards.
+//
+// Population forward at batch=1/member is a batched GEMV: memory-bound on
+// the member's own weights, so the design goal is clean streaming (8-byte
+// bf16x4 / 16-byte float4 loads per lane, G13) + fp32 accumulate.
+// Flat-layout guarantee used here: every layer's weight offset and row
+// stride are multiples of 4 elements (checked in ops/pop_models.py), and
+// population rows are padded to 16B (perturb row_stride), so vector loads
+// stay aligned.
+#pragma once
+#include "common.h"
+
+namespace des {
+
+// load 4 consecutive elements as fp32
+__device__ __forceinline__ float4 load4(const uint16_t* p) {  // bf16x4
+  uint2 pk = *reinterpret_cast<const uint2*>(p);
+  float4 r;
+  r.x = bf16_to_f32((uint16_t)(pk.x & 0xFFFF));
+  r.y = bf16_to_f32((uint16_t)(pk.x >> 16));
+  r.z = bf16_to_f32((uint16_t)(pk.y & 0xFFFF));
+  r.w = bf16_to_f32((uint16_t)(pk.y >> 16));
+  return r;
+}
+__device__ __forceinline__ float4 load4(const float* p) {     // fp32x4
+  return *reinterpret_cast<const float4*>(p);
+}
+__device__ __forceinline__ float load1(const uint16_t* p) { return bf16_to_f32(*p); }
+__device__ __forceinline__ float load1(const float* p) { return *p; }
+
+// y[row] over rows [wid, wid+NW, ...): act(dot(W[row], x) + b[row])
+// W rows at w + row*stride (elements), x fp32 in LDS, y fp32 in LDS.
+// ACT: 0 = identity, 1 = tanh, 2 = relu
+template <typename WT, int ACT>
+__device__ void gemv_rows(const WT* __restrict__ w, const WT* __restrict__ b,
+                          const float* __restrict__ x, float* __restrict__ y,
+                          int in_dim, int out_dim, int64_t stride, int lane,
+                          int wid, int nwaves) {
+  int full = in_dim & ~3;
+  for (int row = wid; row < out_dim; row += nwaves) {
+    const WT* wr = w + (int64_t)row * stride;
+    float acc = 0.f;
+    for (int i = lane * 4; i < full; i += 64 * 4) {
+      float4 wv = load4(wr + i);
+      acc += wv.x * x[i] + wv.y * x[i + 1] + wv.z * x[i + 2] + wv.w * x[i + 3];
+    }
+    for (int i = full + lane; i < in_dim; i += 64) acc += load1(wr + i) * x[i];
+    acc = wave_reduce_sum_f(acc);
+    if (lane == 0) {
+      acc += load1(b + row);
+      if (ACT == 1) acc = tanhf(acc);
+      if (ACT == 2) acc = fmaxf(acc, 0.f);
+      y[row] = acc;
+    }
+  }
+}
+
+}  // namespace des

@@ -23,15 +23,15 @@ __global__ void perturb_kernel(OutT* __restrict__ out,
                                const int64_t* __restrict__ offsets,
                                const float* __restrict__ signs, float sigma,
                                const float* __restrict__ sigma_vec, int64_t P,
-                               int64_t D) {
+                               int64_t D, int64_t out_stride) {
   int64_t i = blockIdx.y;  // member
   if (i >= P) return;
   const float s = signs[i] * (sigma_vec ? 1.0f : sigma);
   const float* src = table + offsets[i];
-  OutT* dst = out + i * D;
-  // row base alignment: odd-D models (e.g. Humanoid MLP D=166673) put row i
-  // at a non-vector boundary; fall back to element stores for those members
-  const bool row_aligned = ((i * D) & 3) == 0;
+  OutT* dst = out + i * out_stride;
+  // rows are normally padded to 16B (out_stride % 8 == 0) so vector stores
+  // stay aligned; unpadded odd-D rows fall back to element stores
+  const bool row_aligned = ((i * out_stride) & 3) == 0;
   int64_t d4 = D >> 2;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < d4;
@@ -133,24 +133,24 @@ extern "C" {
 void des_launch_perturb_f32(float* out, const float* theta, const float* table,
                             const int64_t* offsets, const float* signs,
                             float sigma, const float* sigma_vec, int64_t P,
-                            int64_t D, hipStream_t stream) {
+                            int64_t D, int64_t out_stride, hipStream_t stream) {
   int threads = 256;
   int chunks = (int)std::min<int64_t>(((D >> 2) + threads - 1) / threads, 64);
   hipLaunchKernelGGL(perturb_kernel<float>, dim3(chunks, (uint32_t)P),
                      dim3(threads), 0, stream, out, theta, table, offsets,
-                     signs, sigma, sigma_vec, P, D);
+                     signs, sigma, sigma_vec, P, D, out_stride);
 }
 
 void des_launch_perturb_bf16(uint16_t* out, const float* theta,
                              const float* table, const int64_t* offsets,
                              const float* signs, float sigma,
                              const float* sigma_vec, int64_t P, int64_t D,
-                             hipStream_t stream) {
+                             int64_t out_stride, hipStream_t stream) {
   int threads = 256;
   int chunks = (int)std::min<int64_t>(((D >> 2) + threads - 1) / threads, 64);
   hipLaunchKernelGGL(perturb_kernel<uint16_t>, dim3(chunks, (uint32_t)P),
                      dim3(threads), 0, stream, out, theta, table, offsets,
-                     signs, sigma, sigma_vec, P, D);
+                     signs, sigma, sigma_vec, P, D, out_stride);
 }
 
 void des_launch_wsum(float* acc, float* acc_sq, const float* table,

@@ -1,0 +1,217 @@
+"""HIP rollout engine: the per-generation hot loop on GPU kernels.
+
+Dispatches {env family} x {model family} to the csrc kernels:
+    K7 env step (cartpole / pong / latent-image / humanoid)
+ -> K3/K5 policy forward (MLP / Nature-CNN) with fused normalization + action
+ -> K8 fitness accumulation (fused into env step)
+All state lives in device buffers allocated once; the loop makes zero
+host<->device transfers and no torch ops, so it is hipGraph-capturable
+(engine captures a whole generation's T steps when cfg.use_hip_graphs).
+"""
+from __future__ import annotations
+
+import torch
+
+from ..models import MLPPolicy, NatureCNNPolicy
+from ..models.base import ParamSpec
+
+
+def mlp_layer_offsets(spec: ParamSpec, n_hidden: int):
+    names = []
+    for i in range(n_hidden):
+        names += [f"fc{i+1}.weight", f"fc{i+1}.bias"]
+    names += ["head.weight", "head.bias"]
+    offs = [spec._by_name[n].offset for n in names]
+    for n in names:
+        e = spec._by_name[n]
+        assert e.offset % 4 == 0, (n, e.offset)
+    if n_hidden == 1:
+        offs = offs[:2] + [0, 0] + offs[2:]
+    return offs
+
+
+def nature_cnn_offsets(spec: ParamSpec):
+    names = ["conv1.weight", "conv1.bias", "conv2.weight", "conv2.bias",
+             "conv3.weight", "conv3.bias", "fc.weight", "fc.bias",
+             "head.weight", "head.bias"]
+    offs = [spec._by_name[n].offset for n in names]
+    assert all(o % 4 == 0 for o in offs)
+    return offs
+
+
+class HipRollout:
+    """Owns device buffers + kernel sequencing for one (env, model) pair."""
+
+    SUPPORTED_ENVS = ("cartpole", "pong", "atari57_synth", "procgen_synth",
+                      "humanoid_synth")
+
+    def __init__(self, cfg, env, model, ext, device):
+        self.cfg = cfg
+        self.env_name = cfg.env
+        self.env = env
+        self.model = model
+        self.ext = ext
+        self.device = device
+        self.P = None
+        if isinstance(model, MLPPolicy):
+            self.kind = "mlp"
+            self.h1 = model.hidden_sizes[0]
+            self.h2 = model.hidden_sizes[1] if len(model.hidden_sizes) > 1 else 0
+            assert len(model.hidden_sizes) <= 2, "HIP MLP path supports <=2 hidden"
+            self.offs = mlp_layer_offsets(model.spec, len(model.hidden_sizes))
+        elif isinstance(model, NatureCNNPolicy):
+            self.kind = "nature_cnn"
+            self.offs_t = torch.tensor(nature_cnn_offsets(model.spec),
+                                       dtype=torch.int64, device=device)
+        else:
+            raise NotImplementedError(type(model))
+
+    @classmethod
+    def supports(cls, cfg, model) -> bool:
+        if cfg.env not in cls.SUPPORTED_ENVS:
+            return False
+        if isinstance(model, NatureCNNPolicy):
+            return cfg.precision == "bf16"  # kernel consumes bf16 pop weights
+        return isinstance(model, MLPPolicy) and len(model.hidden_sizes) <= 2
+
+    # ------------------------------------------------------------------ #
+    def _alloc(self, P: int):
+        dev = self.device
+        self.P = P
+        z = lambda *s, dt=torch.float32: torch.zeros(*s, dtype=dt, device=dev)  # noqa: E731
+        self._empty = torch.empty(0, device=dev)
+        self.fitness = z(P)
+        self.alive = z(P, dt=torch.uint8)
+        self.ep_len = z(P)
+        self.actions_i = z(P, dt=torch.int32)
+        self.actions_f = (z(P, self.env.num_actions)
+                          if self.env.continuous else None)
+        e = self.env_name
+        if e == "cartpole":
+            self.st_state = z(P, 4)
+            self.st_steps = z(P, dt=torch.int32)
+            self.st_rctr = z(P, dt=torch.int64)
+        elif e == "pong":
+            for n in ("bx", "by", "vx", "vy", "py", "oy"):
+                setattr(self, "pg_" + n, z(P))
+            self.pg_score = z(P, dt=torch.int32)
+            self.pg_points = z(P, dt=torch.int32)
+            self.pg_serve = z(P, dt=torch.int64)
+            self.frames = z(P, 4, 84, 84, dt=torch.uint8)
+            # phase p (= t % 4): logical channel c -> physical slot (p+c) % 4
+            self.slot_maps = [torch.tensor([(p + c) % 4 for c in range(4)],
+                                           dtype=torch.int32, device=dev)
+                              for p in range(4)]
+        elif e in ("atari57_synth", "procgen_synth"):
+            self.sy_z = z(P, 8)
+            self.sy_steps = z(P, dt=torch.int32)
+            c, hw, _ = self.env.obs_shape
+            self.obs_bf16 = z(P, c, hw, hw, dt=torch.bfloat16)
+            self.slot_map = torch.tensor([0, 1, 2, 3], dtype=torch.int32,
+                                         device=dev)
+        elif e == "humanoid_synth":
+            self.hu_s = z(P, 376)
+            self.hu_steps = z(P, dt=torch.int32)
+
+    # ------------------------------------------------------------------ #
+    def _reset(self, gen_seed: int, env_ids: torch.Tensor):
+        e = self.env_name
+        ids = env_ids.to(self.device)
+        if e == "cartpole":
+            self.ext.cartpole_reset(self.st_state, self.st_steps, self.st_rctr,
+                                    self.fitness, self.alive, self.ep_len,
+                                    ids, gen_seed * 0x10001)
+        elif e == "pong":
+            self.ext.pong_reset(self.pg_bx, self.pg_by, self.pg_vx, self.pg_vy,
+                                self.pg_py, self.pg_oy, self.pg_score,
+                                self.pg_points, self.pg_serve, self.frames,
+                                self.fitness, self.alive, self.ep_len, ids,
+                                gen_seed)
+        elif e in ("atari57_synth", "procgen_synth"):
+            self.ext.latent_reset(self.sy_z, self.sy_steps, self.fitness,
+                                  self.alive, self.ep_len, ids, gen_seed)
+            self.ext.latent_render(self.sy_z, self.obs_bf16)
+        elif e == "humanoid_synth":
+            self.ext.humanoid_reset(self.hu_s, self.hu_steps, self.fitness,
+                                    self.alive, self.ep_len, ids, gen_seed)
+        self._ids = ids
+        self._gen_seed = gen_seed
+
+    # ------------------------------------------------------------------ #
+    def _forward(self, pop_w: torch.Tensor, normalizer, slot_newest: int):
+        """One policy forward: obs buffers -> action buffers."""
+        if self.kind == "mlp":
+            obs = self.st_state if self.env_name == "cartpole" else self.hu_s
+            nmean = getattr(normalizer, "_mean", None)
+            if nmean is not None and nmean.device != self.device:
+                nmean = None
+            if nmean is not None:
+                ninv = normalizer._inv_std
+                clip = normalizer.clip
+            else:
+                nmean = torch.empty(0, device=self.device)
+                ninv = nmean
+                clip = 0.0
+            act = self.actions_f if self.env.continuous else self.actions_i
+            self.ext.mlp_forward(pop_w, obs, nmean, ninv, clip, act,
+                                 self._empty, self.h1, self.h2,
+                                 self.env.num_actions, self.offs)
+        else:  # nature_cnn
+            if self.env_name == "pong":
+                frames = self.frames
+                sm = self.slot_maps[(slot_newest + 1) % 4]
+            else:
+                frames = self.obs_bf16
+                sm = self.slot_map
+            self.ext.nature_cnn_forward(pop_w, frames, sm, self.actions_i,
+                                        self._empty, self.env.num_actions,
+                                        self.offs_t)
+
+    def _env_step(self, slot: int):
+        e = self.env_name
+        if e == "cartpole":
+            self.ext.cartpole_step(self.st_state, self.st_steps, self.st_rctr,
+                                   self.actions_i, self.fitness, self.alive,
+                                   self.ep_len, self._ids,
+                                   self._gen_seed * 0x10001)
+        elif e == "pong":
+            self.ext.pong_step(self.pg_bx, self.pg_by, self.pg_vx, self.pg_vy,
+                               self.pg_py, self.pg_oy, self.pg_score,
+                               self.pg_points, self.pg_serve, self.frames,
+                               slot, self.actions_i, self.fitness, self.alive,
+                               self.ep_len, self._ids, self._gen_seed)
+        elif e in ("atari57_synth", "procgen_synth"):
+            self.ext.latent_step(self.sy_z, self.sy_steps, self.actions_i,
+                                 self.fitness, self.alive, self.ep_len,
+                                 self.env.num_actions, self.env.horizon)
+            self.ext.latent_render(self.sy_z, self.obs_bf16)
+        elif e == "humanoid_synth":
+            self.ext.humanoid_step(self.hu_s, self.hu_steps, self.actions_f,
+                                   self.fitness, self.alive, self.ep_len,
+                                   self.env.horizon)
+
+    def _observe(self, normalizer):
+        """K12: fold current obs into the normalizer's fp64 delta buffers."""
+        if not hasattr(normalizer, "_dsum"):
+            return
+        obs = self.st_state if self.env_name == "cartpole" else self.hu_s
+        self.ext.normalizer_observe(obs, normalizer._dsum, normalizer._dsumsq)
+        normalizer._dcount += obs.shape[0]
+
+    # ------------------------------------------------------------------ #
+    def run(self, pop_w: torch.Tensor, gen_seed: int, env_ids: torch.Tensor,
+            episode_steps: int, normalizer):
+        P = pop_w.shape[0]
+        if self.P != P:
+            self._alloc(P)
+        is_state_env = len(self.env.obs_shape) == 1
+        self._reset(gen_seed, env_ids)
+        # pong ring: reset wrote the initial frame to every slot; step t
+        # writes slot t % 4 and the forward sees slots in logical order
+        for t in range(episode_steps):
+            slot = t % 4
+            if is_state_env:
+                self._observe(normalizer)
+            self._forward(pop_w, normalizer, slot_newest=(t - 1) % 4)
+            self._env_step(slot)
+        return self.fitness.clone(), P * episode_steps, float(self.ep_len.mean())
