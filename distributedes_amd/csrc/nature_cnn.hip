@@ -1,20 +1,19 @@
-// K5: conv_forward_pop (Nature-CNN) — whole per-member policy forward in ONE
-// block-per-member kernel.
+// K5: conv_forward_pop (Nature-CNN) — per-member policy forward as three
+// kernels sized for their regimes:
 //
-// Per member (batch=1): conv 8x8s4x32 -> 4x4s2x64 -> 3x3s1x64 -> fc 3136->512
-// -> head 512->A -> argmax.  Design for MI355X:
-//   * one 512-thread block (8 waves) per member; activations live entirely in
-//     LDS (~123 KiB carve, 1 block/CU) and never touch HBM;
-//   * per-stage weight tiles are staged global->LDS with coalesced reads and
-//     a conflict-free [k/4][oc][4] layout so the inner loop is ds_read_b64 of
-//     4 consecutive bf16 per lane (G13/G4);
-//   * conv inner loops: activation value broadcast (uniform LDS address),
-//     per-lane weight vector, fp32 accumulate;
-//   * the FC layer (94% of weights -> the HBM-bound stage) streams rows with
-//     the wave-GEMV (gemv.h): bf16x4 loads, 8 waves in flight per CU;
-//   * action argmax fused into the epilogue (one kernel per env step).
-// Step time at pop=8192/GPU is bounded by FC weight traffic:
-// 8192 * 3.2 MiB / 6.3 TBps ~= 4.3 ms; conv stages are VALU work (~0.8 ms).
+//   1. nature_conv_kernel   (block/member, 512 thr, ~105 KiB LDS)
+//      conv 8x8s4x32 -> 4x4s2x64 -> 3x3s1x64; activations in LDS; per-stage
+//      weight tiles staged coalesced into a conflict-free [k/4][oc][4] bf16
+//      layout; separable LDS offset tables kill per-MAC index math; conv1
+//      runs two 16-channel passes to stay under the VGPR cap (no spills).
+//      Writes the flattened conv output (fc input, [P,3136] bf16) to HBM.
+//   2. nature_fc_kernel     (grid P x 8 row-groups, 256 thr, ~7 KiB LDS)
+//      the HBM-bound stage (94% of weights): each wave streams 4 rows
+//      concurrently with bf16x8 (16 B) loads -> 4 independent load streams
+//      per wave x many blocks/CU saturate HBM (~3.2 MiB/member/step).
+//   3. nature_head_kernel   (block/member) 512->A + argmax (+ logits).
+//
+// Step floor at pop=8192/GPU: FC traffic 8192*3.2 MiB / 6.3 TBps ~= 4.3 ms.
 #include "common.h"
 #include "gemv.h"
 
@@ -24,20 +23,20 @@ namespace {
 constexpr int C_IN = 4, HW = 84;
 constexpr int C1 = 32, O1 = 20, K1 = 8, S1 = 4;   // conv1
 constexpr int C2 = 64, O2 = 9, K2 = 4, S2 = 2;    // conv2
-constexpr int C3 = 64, O3 = 7, K3 = 3;            // conv3 (s1)
+constexpr int C3 = 64, O3 = 7;                    // conv3 (3x3 s1)
 constexpr int FC_IN = C3 * O3 * O3;               // 3136
 constexpr int FC_OUT = 512;
 constexpr int MAX_A = 32;
 
+// ---------------- kernel 1: convs ----------------
 // LDS carve (bytes, 16-aligned)
-constexpr int W_ELEMS = 36864;                         // max transposed tile
+constexpr int W_ELEMS = C2 * 512;                      // 32768 (conv2 tile)
 constexpr int OFF_W = 0;                               // bf16 [W_ELEMS]
 constexpr int OFF_A1 = OFF_W + W_ELEMS * 2;            // bf16 [32*20*20]
 constexpr int OFF_A2 = OFF_A1 + C1 * O1 * O1 * 2;      // bf16 [64*9*9]
-constexpr int OFF_A3 = OFF_A2 + ((C2 * O2 * O2 * 2 + 15) & ~15);  // f32 [3136]
-constexpr int OFF_FC = OFF_A3 + FC_IN * 4;             // f32 [512]
-constexpr int OFF_HD = OFF_FC + FC_OUT * 4;            // f32 [MAX_A]
-constexpr int LDS_BYTES = OFF_HD + MAX_A * 4;
+constexpr int OFF_T2 = OFF_A2 + ((C2 * O2 * O2 * 2 + 15) & ~15);  // int [512]
+constexpr int OFF_T3 = OFF_T2 + 512 * 4;               // int [576]
+constexpr int CONV_LDS = OFF_T3 + 576 * 4;
 
 template <typename IT>
 __device__ __forceinline__ float obs_val(IT v);
@@ -49,19 +48,18 @@ __device__ __forceinline__ float obs_val<uint16_t>(uint16_t v) {
 }
 
 template <typename IT>
-__global__ __launch_bounds__(512, 1) void nature_cnn_forward_kernel(
+__global__ __launch_bounds__(512, 2) void nature_conv_kernel(
     const uint16_t* __restrict__ popw, int64_t row_stride,
-    const IT* __restrict__ frames,  // [P][4][84][84], channel c = slot[c]
+    const IT* __restrict__ frames,     // [P][4][84][84]
     const int* __restrict__ slot_map,  // [4] logical->physical slot
-    int* __restrict__ actions, float* __restrict__ logits_out, int A, int P,
-    const int64_t* __restrict__ offs) {  // ow1,ob1,ow2,ob2,ow3,ob3,owf,obf,owh,obh
+    uint16_t* __restrict__ fc_in,      // out: [P][3136] bf16
+    int P, const int64_t* __restrict__ offs) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint16_t* w_lds = reinterpret_cast<uint16_t*>(smem + OFF_W);
   uint16_t* a1 = reinterpret_cast<uint16_t*>(smem + OFF_A1);
   uint16_t* a2 = reinterpret_cast<uint16_t*>(smem + OFF_A2);
-  float* a3 = reinterpret_cast<float*>(smem + OFF_A3);
-  float* fcb = reinterpret_cast<float*>(smem + OFF_FC);
-  float* hd = reinterpret_cast<float*>(smem + OFF_HD);
+  int* t2 = reinterpret_cast<int*>(smem + OFF_T2);
+  int* t3 = reinterpret_cast<int*>(smem + OFF_T3);
 
   const int i = blockIdx.x;
   if (i >= P) return;
@@ -71,7 +69,16 @@ __global__ __launch_bounds__(512, 1) void nature_cnn_forward_kernel(
   const int s0 = slot_map[0], s1m = slot_map[1], s2m = slot_map[2],
             s3 = slot_map[3];
 
-  // ---- stage W1: transpose conv1 weights [oc][icc] -> [icc][oc] ----------
+  // separable activation-offset tables: a1 index = base(pos) + t2[icc]
+  for (int e = tid; e < 512; e += 512) {
+    int ic = e >> 4, ky = (e >> 2) & 3, kx = e & 3;
+    t2[e] = ic * (O1 * O1) + ky * O1 + kx;
+  }
+  for (int e = tid; e < 576; e += 512) {
+    int ic = e / 9, r = e % 9;
+    t3[e] = ic * (O2 * O2) + (r / 3) * O2 + (r % 3);
+  }
+  // stage conv1 weights transposed [icc][oc] (8 KiB of the w buffer)
   {
     const uint16_t* w1 = w + offs[0];
     for (int e = tid; e < C1 * 256; e += 512) {
@@ -81,48 +88,50 @@ __global__ __launch_bounds__(512, 1) void nature_cnn_forward_kernel(
   }
   __syncthreads();
 
-  // ---- conv1: thread = output position (400), 32 oc accumulators ---------
+  // ---- conv1: thread = position (400 of 512), two passes of 16 oc ------
   {
-    float acc[C1];
-    const int pos = tid;  // 0..399 active
-    if (pos < O1 * O1) {
+    const int pos = tid;
+    const int oy = pos / O1, ox = pos % O1;
+    const uint16_t* b1 = w + offs[1];
+    for (int half = 0; half < 2; ++half) {
+      float acc[16];
+      if (pos < O1 * O1) {
 #pragma unroll
-      for (int oc = 0; oc < C1; ++oc) acc[oc] = 0.f;
-      const int oy = pos / O1, ox = pos % O1;
-      for (int ic = 0; ic < C_IN; ++ic) {
-        const int slot = (ic == 0 ? s0 : ic == 1 ? s1m : ic == 2 ? s2m : s3);
-        const IT* plane = obs + slot * (HW * HW);
-        for (int ky = 0; ky < K1; ++ky) {
-          const IT* row = plane + (oy * S1 + ky) * HW + ox * S1;
-          float px[K1];
+        for (int k = 0; k < 16; ++k) acc[k] = 0.f;
+        for (int ic = 0; ic < C_IN; ++ic) {
+          const int slot = (ic == 0 ? s0 : ic == 1 ? s1m : ic == 2 ? s2m : s3);
+          const IT* plane = obs + slot * (HW * HW);
+          for (int ky = 0; ky < K1; ++ky) {
+            const IT* row = plane + (oy * S1 + ky) * HW + ox * S1;
+            const uint16_t* wrow =
+                w_lds + ((ic * K1 + ky) * K1) * C1 + half * 16;
 #pragma unroll
-          for (int kx = 0; kx < K1; ++kx) px[kx] = obs_val<IT>(row[kx]);
-          const uint16_t* wrow = w_lds + ((ic * K1 + ky) * K1) * C1;
+            for (int kx = 0; kx < K1; ++kx) {
+              float px = obs_val<IT>(row[kx]);
+              const uint16_t* wv = wrow + kx * C1;
 #pragma unroll
-          for (int kx = 0; kx < K1; ++kx) {
-            const uint16_t* wv = wrow + kx * C1;
-#pragma unroll  // static acc[] indices (rule 20: runtime-indexed regs spill)
-            for (int oc = 0; oc < C1; oc += 4) {
-              float4 w4 = des::load4(wv + oc);
-              acc[oc] += px[kx] * w4.x;
-              acc[oc + 1] += px[kx] * w4.y;
-              acc[oc + 2] += px[kx] * w4.z;
-              acc[oc + 3] += px[kx] * w4.w;
+              for (int k = 0; k < 16; k += 4) {
+                float4 w4 = des::load4(wv + k);
+                acc[k] += px * w4.x;
+                acc[k + 1] += px * w4.y;
+                acc[k + 2] += px * w4.z;
+                acc[k + 3] += px * w4.w;
+              }
             }
           }
         }
-      }
-      const uint16_t* b1 = w + offs[1];
 #pragma unroll
-      for (int oc = 0; oc < C1; ++oc) {
-        float v = fmaxf(acc[oc] + des::load1(b1 + oc), 0.f);
-        a1[oc * (O1 * O1) + pos] = des::f32_to_bf16(v);
+        for (int k = 0; k < 16; ++k) {
+          int oc = half * 16 + k;
+          float v = fmaxf(acc[k] + des::load1(b1 + oc), 0.f);
+          a1[oc * (O1 * O1) + pos] = des::f32_to_bf16(v);
+        }
       }
     }
   }
   __syncthreads();
 
-  // ---- stage W2: [oc][icc](K=512) -> [icc/4][oc][4] -----------------------
+  // ---- stage W2: [oc][icc](K=512) -> [icc/4][oc][4] ---------------------
   {
     const uint16_t* w2 = w + offs[2];
     for (int e = tid; e < C2 * 512; e += 512) {
@@ -132,73 +141,165 @@ __global__ __launch_bounds__(512, 1) void nature_cnn_forward_kernel(
   }
   __syncthreads();
 
-  // ---- conv2: lane = oc (64), wave = position slice (81 positions) --------
+  // ---- conv2: lane = oc; 8 positions per wave iteration in registers ----
   {
     const uint16_t* b2 = w + offs[3];
     const float bias = des::load1(b2 + lane);
-    for (int pos = wid; pos < O2 * O2; pos += 8) {
-      const int oy = pos / O2, ox = pos % O2;
-      float acc = 0.f;
+    for (int pb = wid * 8; pb < O2 * O2; pb += 64) {
+      float acc[8];
+      int pbase[8];
+#pragma unroll
+      for (int p = 0; p < 8; ++p) {
+        acc[p] = 0.f;
+        int pos = pb + p;
+        if (pos >= O2 * O2) pos = O2 * O2 - 1;  // clamp; result discarded
+        pbase[p] = (pos / O2) * S2 * O1 + (pos % O2) * S2;
+      }
       for (int icc = 0; icc < 512; icc += 4) {
         float4 w4 = des::load4(w_lds + (icc >> 2) * (C2 * 4) + lane * 4);
-        // icc -> (ic, ky, kx): ic = icc/16, ky = (icc%16)/4, kx = icc%4
-        float av[4];
+        const int4 off = *reinterpret_cast<const int4*>(t2 + icc);
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          int cc = icc + j;
-          int ic = cc >> 4, ky = (cc >> 2) & 3, kx = cc & 3;
-          av[j] = des::bf16_to_f32(
-              a1[ic * (O1 * O1) + (oy * S2 + ky) * O1 + (ox * S2 + kx)]);
+        for (int p = 0; p < 8; ++p) {
+          float a0 = des::bf16_to_f32(a1[pbase[p] + off.x]);
+          float a1v = des::bf16_to_f32(a1[pbase[p] + off.y]);
+          float a2v = des::bf16_to_f32(a1[pbase[p] + off.z]);
+          float a3v = des::bf16_to_f32(a1[pbase[p] + off.w]);
+          acc[p] += a0 * w4.x + a1v * w4.y + a2v * w4.z + a3v * w4.w;
         }
-        acc += av[0] * w4.x + av[1] * w4.y + av[2] * w4.z + av[3] * w4.w;
       }
-      a2[lane * (O2 * O2) + pos] = des::f32_to_bf16(fmaxf(acc + bias, 0.f));
+#pragma unroll
+      for (int p = 0; p < 8; ++p) {
+        int pos = pb + p;
+        if (pos < O2 * O2)
+          a2[lane * (O2 * O2) + pos] =
+              des::f32_to_bf16(fmaxf(acc[p] + bias, 0.f));
+      }
     }
   }
   __syncthreads();
 
-  // ---- stage W3: [oc][icc](K=576) -> [icc/4][oc][4] -----------------------
+  // ---- conv3: two oc-halves (tile 36 KiB > W buffer? no: 32 oc = 18 KiB)
   {
     const uint16_t* w3 = w + offs[4];
-    for (int e = tid; e < C3 * 576; e += 512) {
-      int oc = e / 576, icc = e % 576;
-      w_lds[(icc >> 2) * (C3 * 4) + oc * 4 + (icc & 3)] = w3[e];
-    }
-  }
-  __syncthreads();
-
-  // ---- conv3: lane = oc (64), 49 positions --------------------------------
-  {
     const uint16_t* b3 = w + offs[5];
-    const float bias = des::load1(b3 + lane);
-    for (int pos = wid; pos < O3 * O3; pos += 8) {
-      const int oy = pos / O3, ox = pos % O3;
-      float acc = 0.f;
-      for (int icc = 0; icc < 576; icc += 4) {
-        float4 w4 = des::load4(w_lds + (icc >> 2) * (C3 * 4) + lane * 4);
-        float av[4];
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          int cc = icc + j;
-          int ic = cc / 9, r = cc % 9, ky = r / 3, kx = r % 3;
-          av[j] = des::bf16_to_f32(
-              a2[ic * (O2 * O2) + (oy + ky) * O2 + (ox + kx)]);
-        }
-        acc += av[0] * w4.x + av[1] * w4.y + av[2] * w4.z + av[3] * w4.w;
+    uint16_t* out = fc_in + (int64_t)i * FC_IN;
+    for (int half = 0; half < 2; ++half) {
+      // stage [icc/4][oc32][4] for ocs [half*32, half*32+32)
+      for (int e = tid; e < 32 * 576; e += 512) {
+        int oc = e / 576, icc = e % 576;
+        w_lds[(icc >> 2) * 128 + oc * 4 + (icc & 3)] =
+            w3[(half * 32 + oc) * 576 + icc];
       }
-      a3[lane * (O3 * O3) + pos] = fmaxf(acc + bias, 0.f);
+      __syncthreads();
+      // lane&31 = oc-in-half; lane>>5 selects position stream (2 streams)
+      const int oc = (lane & 31);
+      const float bias = des::load1(b3 + half * 32 + oc);
+      for (int pb = (wid * 2 + (lane >> 5)) * 4; pb < O3 * O3; pb += 64) {
+        float acc[4];
+        int pbase[4];
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+          acc[p] = 0.f;
+          int pos = pb + p;
+          if (pos >= O3 * O3) pos = O3 * O3 - 1;
+          pbase[p] = (pos / O3) * O2 + (pos % O3);
+        }
+        for (int icc = 0; icc < 576; icc += 4) {
+          float4 w4 = des::load4(w_lds + (icc >> 2) * 128 + oc * 4);
+          const int4 off = *reinterpret_cast<const int4*>(t3 + icc);
+#pragma unroll
+          for (int p = 0; p < 4; ++p) {
+            float a0 = des::bf16_to_f32(a2[pbase[p] + off.x]);
+            float b0 = des::bf16_to_f32(a2[pbase[p] + off.y]);
+            float c0 = des::bf16_to_f32(a2[pbase[p] + off.z]);
+            float d0 = des::bf16_to_f32(a2[pbase[p] + off.w]);
+            acc[p] += a0 * w4.x + b0 * w4.y + c0 * w4.z + d0 * w4.w;
+          }
+        }
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+          int pos = pb + p;
+          if (pos < O3 * O3)
+            out[(half * 32 + oc) * (O3 * O3) + pos] =
+                des::f32_to_bf16(fmaxf(acc[p] + bias, 0.f));
+        }
+      }
+      __syncthreads();
     }
   }
+}
+
+// ---------------- kernel 2: fc 3136 -> 512 (relu) ----------------
+// grid (P, 8): block handles 64 rows of one member; 4 waves x 4 rows
+// concurrently; 16-byte bf16x8 weight loads; x staged bf16 in LDS.
+__global__ __launch_bounds__(256) void nature_fc_kernel(
+    const uint16_t* __restrict__ popw, int64_t row_stride,
+    const uint16_t* __restrict__ fc_in,  // [P][3136]
+    float* __restrict__ fc_out,          // [P][512]
+    int P, int64_t wf_off, int64_t bf_off) {
+  __shared__ uint16_t xs[FC_IN];
+  const int i = blockIdx.x;
+  if (i >= P) return;
+  const int rg = blockIdx.y;           // row group: 64 rows
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  const uint16_t* x = fc_in + (int64_t)i * FC_IN;
+  for (int e = tid * 8; e < FC_IN; e += 256 * 8)
+    *reinterpret_cast<uint4*>(xs + e) =
+        *reinterpret_cast<const uint4*>(x + e);
   __syncthreads();
 
-  // ---- fc 3136 -> 512 (relu): the HBM-bound stage -------------------------
-  des::gemv_rows<uint16_t, 2>(w + offs[6], w + offs[7], a3, fcb, FC_IN, FC_OUT,
-                              FC_IN, lane, wid, 8);
-  __syncthreads();
+  const uint16_t* wf = popw + (int64_t)i * row_stride + wf_off;
+  const uint16_t* bf = popw + (int64_t)i * row_stride + bf_off;
+  // wave wid covers rows rg*64 + wid*16 + {0..15}, 4 at a time
+  for (int rr = 0; rr < 16; rr += 4) {
+    const int row0 = rg * 64 + wid * 16 + rr;
+    const uint16_t* w0 = wf + (int64_t)row0 * FC_IN;
+    float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
+    for (int ib = lane * 8; ib < FC_IN; ib += 64 * 8) {
+      // x chunk (conflict-free b128: lane l -> banks 4l..4l+3 mod 64)
+      uint4 xp = *reinterpret_cast<const uint4*>(xs + ib);
+      const uint16_t* xc = reinterpret_cast<const uint16_t*>(&xp);
+      uint4 wa = *reinterpret_cast<const uint4*>(w0 + ib);
+      uint4 wb = *reinterpret_cast<const uint4*>(w0 + FC_IN + ib);
+      uint4 wc = *reinterpret_cast<const uint4*>(w0 + 2 * FC_IN + ib);
+      uint4 wd = *reinterpret_cast<const uint4*>(w0 + 3 * FC_IN + ib);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xv = des::bf16_to_f32(xc[j]);
+        acc0 += xv * des::bf16_to_f32(reinterpret_cast<const uint16_t*>(&wa)[j]);
+        acc1 += xv * des::bf16_to_f32(reinterpret_cast<const uint16_t*>(&wb)[j]);
+        acc2 += xv * des::bf16_to_f32(reinterpret_cast<const uint16_t*>(&wc)[j]);
+        acc3 += xv * des::bf16_to_f32(reinterpret_cast<const uint16_t*>(&wd)[j]);
+      }
+    }
+    acc0 = des::wave_reduce_sum_f(acc0);
+    acc1 = des::wave_reduce_sum_f(acc1);
+    acc2 = des::wave_reduce_sum_f(acc2);
+    acc3 = des::wave_reduce_sum_f(acc3);
+    if (lane < 4) {
+      float acc = lane == 0 ? acc0 : lane == 1 ? acc1 : lane == 2 ? acc2 : acc3;
+      float bias = des::bf16_to_f32(bf[row0 + lane]);
+      fc_out[(int64_t)i * FC_OUT + row0 + lane] = fmaxf(acc + bias, 0.f);
+    }
+  }
+}
 
-  // ---- head 512 -> A + argmax --------------------------------------------
-  des::gemv_rows<uint16_t, 0>(w + offs[8], w + offs[9], fcb, hd, FC_OUT, A,
-                              FC_OUT, lane, wid, 8);
+// ---------------- kernel 3: head 512 -> A + argmax ----------------
+__global__ __launch_bounds__(256) void nature_head_kernel(
+    const uint16_t* __restrict__ popw, int64_t row_stride,
+    const float* __restrict__ fc_out, int* __restrict__ actions,
+    float* __restrict__ logits_out, int A, int P, int64_t wh_off,
+    int64_t bh_off) {
+  __shared__ float xs[FC_OUT];
+  __shared__ float hd[MAX_A];
+  const int i = blockIdx.x;
+  if (i >= P) return;
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  for (int e = tid; e < FC_OUT; e += 256) xs[e] = fc_out[(int64_t)i * FC_OUT + e];
+  __syncthreads();
+  const uint16_t* w = popw + (int64_t)i * row_stride;
+  des::gemv_rows<uint16_t, 0>(w + wh_off, w + bh_off, xs, hd, FC_OUT, A,
+                              FC_OUT, lane, wid, 4);
   __syncthreads();
   if (logits_out && tid < A) logits_out[(int64_t)i * A + tid] = hd[tid];
   if (tid == 0) {
@@ -214,26 +315,31 @@ __global__ __launch_bounds__(512, 1) void nature_cnn_forward_kernel(
 
 extern "C" void des_launch_nature_cnn_forward(
     const uint16_t* popw, int64_t row_stride, const void* frames, bool obs_u8,
-    const int* slot_map_dev, int* actions, float* logits_out, int A, int P,
-    const int64_t* offs_dev, hipStream_t stream) {
-  // dynamic-LDS carve is 121.5 KiB (> the 64 KiB default cap): opt in once
+    const int* slot_map_dev, uint16_t* fc_in, float* fc_out, int* actions,
+    float* logits_out, int A, int P, const int64_t* offs_dev,
+    const int64_t* offs_host, hipStream_t stream) {
   static bool attr_set = false;
   if (!attr_set) {
-    hipFuncSetAttribute((const void*)nature_cnn_forward_kernel<uint8_t>,
-                        hipFuncAttributeMaxDynamicSharedMemorySize, LDS_BYTES);
-    hipFuncSetAttribute((const void*)nature_cnn_forward_kernel<uint16_t>,
-                        hipFuncAttributeMaxDynamicSharedMemorySize, LDS_BYTES);
+    hipFuncSetAttribute((const void*)nature_conv_kernel<uint8_t>,
+                        hipFuncAttributeMaxDynamicSharedMemorySize, CONV_LDS);
+    hipFuncSetAttribute((const void*)nature_conv_kernel<uint16_t>,
+                        hipFuncAttributeMaxDynamicSharedMemorySize, CONV_LDS);
     attr_set = true;
   }
-  dim3 grid((uint32_t)P), block(512);
   if (obs_u8)
-    hipLaunchKernelGGL(nature_cnn_forward_kernel<uint8_t>, grid, block,
-                       LDS_BYTES, stream, popw, row_stride,
-                       (const uint8_t*)frames, slot_map_dev, actions,
-                       logits_out, A, P, offs_dev);
+    hipLaunchKernelGGL(nature_conv_kernel<uint8_t>, dim3((uint32_t)P),
+                       dim3(512), CONV_LDS, stream, popw, row_stride,
+                       (const uint8_t*)frames, slot_map_dev, fc_in, P,
+                       offs_dev);
   else
-    hipLaunchKernelGGL(nature_cnn_forward_kernel<uint16_t>, grid, block,
-                       LDS_BYTES, stream, popw, row_stride,
-                       (const uint16_t*)frames, slot_map_dev, actions,
-                       logits_out, A, P, offs_dev);
+    hipLaunchKernelGGL(nature_conv_kernel<uint16_t>, dim3((uint32_t)P),
+                       dim3(512), CONV_LDS, stream, popw, row_stride,
+                       (const uint16_t*)frames, slot_map_dev, fc_in, P,
+                       offs_dev);
+  hipLaunchKernelGGL(nature_fc_kernel, dim3((uint32_t)P, 8), dim3(256), 0,
+                     stream, popw, row_stride, fc_in, fc_out, P, offs_host[6],
+                     offs_host[7]);
+  hipLaunchKernelGGL(nature_head_kernel, dim3((uint32_t)P), dim3(256), 0,
+                     stream, popw, row_stride, fc_out, actions, logits_out, A,
+                     P, offs_host[8], offs_host[9]);
 }

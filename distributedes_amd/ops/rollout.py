@@ -61,8 +61,9 @@ class HipRollout:
             self.offs = mlp_layer_offsets(model.spec, len(model.hidden_sizes))
         elif isinstance(model, NatureCNNPolicy):
             self.kind = "nature_cnn"
-            self.offs_t = torch.tensor(nature_cnn_offsets(model.spec),
-                                       dtype=torch.int64, device=device)
+            self.offs_h = nature_cnn_offsets(model.spec)
+            self.offs_t = torch.tensor(self.offs_h, dtype=torch.int64,
+                                       device=device)
         else:
             raise NotImplementedError(type(model))
 
@@ -80,6 +81,9 @@ class HipRollout:
         self.P = P
         z = lambda *s, dt=torch.float32: torch.zeros(*s, dtype=dt, device=dev)  # noqa: E731
         self._empty = torch.empty(0, device=dev)
+        if self.kind == "nature_cnn":
+            self.fc_in = z(P, 3136, dt=torch.bfloat16)
+            self.fc_out = z(P, 512)
         self.fitness = z(P)
         self.alive = z(P, dt=torch.uint8)
         self.ep_len = z(P)
@@ -163,9 +167,10 @@ class HipRollout:
             else:
                 frames = self.obs_bf16
                 sm = self.slot_map
-            self.ext.nature_cnn_forward(pop_w, frames, sm, self.actions_i,
+            self.ext.nature_cnn_forward(pop_w, frames, sm, self.fc_in,
+                                        self.fc_out, self.actions_i,
                                         self._empty, self.env.num_actions,
-                                        self.offs_t)
+                                        self.offs_t, self.offs_h)
 
     def _env_step(self, slot: int):
         e = self.env_name
