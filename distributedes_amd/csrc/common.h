@@ -110,6 +110,20 @@ __device__ __forceinline__ float bf16_to_f32(uint16_t h) {
   return __uint_as_float(((uint32_t)h) << 16);
 }
 
+// packed bf16 pair + v_dot2c_f32_bf16 (2 MACs/instruction, no unpack)
+typedef __attribute__((ext_vector_type(2))) short bf16x2;
+
+__device__ __forceinline__ float dot2(bf16x2 a, bf16x2 b, float c) {
+  return __builtin_amdgcn_fdot2_f32_bf16(a, b, c, false);
+}
+
+__device__ __forceinline__ bf16x2 pack_bf16(float lo, float hi) {
+  bf16x2 r;
+  r.x = (short)f32_to_bf16(lo);
+  r.y = (short)f32_to_bf16(hi);
+  return r;
+}
+
 }  // namespace des
 
 #define HIP_CHECK(expr)                                                  \

@@ -79,54 +79,71 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
     int ic = e / 9, r = e % 9;
     t3[e] = ic * (O2 * O2) + (r / 3) * O2 + (r % 3);
   }
-  // stage conv1 weights transposed [icc][oc] (8 KiB of the w buffer)
+  // stage conv1 weights as [ic][ky][oc][kx8]: per-oc b128 gives 4 packed
+  // kx-pairs feeding v_dot2c_f32_bf16 directly
   {
     const uint16_t* w1 = w + offs[0];
     for (int e = tid; e < C1 * 256; e += 512) {
-      int oc = e >> 8, icc = e & 255;
-      w_lds[icc * C1 + oc] = w1[e];
+      int oc = e >> 8, rem = e & 255, icky = rem >> 3, kx = e & 7;
+      w_lds[(icky * C1 + oc) * K1 + kx] = w1[e];
     }
   }
   __syncthreads();
 
-  // ---- conv1: thread = position (400 of 512), two passes of 16 oc ------
+  // ---- conv1: thread = position (400 of 512); packed-pair dot2 ---------
   {
     const int pos = tid;
     const int oy = pos / O1, ox = pos % O1;
     const uint16_t* b1 = w + offs[1];
-    for (int half = 0; half < 2; ++half) {
-      float acc[16];
-      if (pos < O1 * O1) {
+    float acc[C1];
+    if (pos < O1 * O1) {
 #pragma unroll
-        for (int k = 0; k < 16; ++k) acc[k] = 0.f;
-        for (int ic = 0; ic < C_IN; ++ic) {
-          const int slot = (ic == 0 ? s0 : ic == 1 ? s1m : ic == 2 ? s2m : s3);
-          const IT* plane = obs + slot * (HW * HW);
-          for (int ky = 0; ky < K1; ++ky) {
-            const IT* row = plane + (oy * S1 + ky) * HW + ox * S1;
-            const uint16_t* wrow =
-                w_lds + ((ic * K1 + ky) * K1) * C1 + half * 16;
+      for (int k = 0; k < C1; ++k) acc[k] = 0.f;
+      for (int ic = 0; ic < C_IN; ++ic) {
+        const int slot = (ic == 0 ? s0 : ic == 1 ? s1m : ic == 2 ? s2m : s3);
+        const IT* plane = obs + slot * (HW * HW);
+        for (int ky = 0; ky < K1; ++ky) {
+          const IT* row = plane + (oy * S1 + ky) * HW + ox * S1;
+          // 8 pixels -> 4 packed bf16 pairs
+          des::bf16x2 pxp[4];
+          if constexpr (sizeof(IT) == 1) {           // u8 frames (dword loads)
+            uint32_t d0 = *reinterpret_cast<const uint32_t*>(row);
+            uint32_t d1 = *reinterpret_cast<const uint32_t*>(row + 4);
 #pragma unroll
-            for (int kx = 0; kx < K1; ++kx) {
-              float px = obs_val<IT>(row[kx]);
-              const uint16_t* wv = wrow + kx * C1;
-#pragma unroll
-              for (int k = 0; k < 16; k += 4) {
-                float4 w4 = des::load4(wv + k);
-                acc[k] += px * w4.x;
-                acc[k + 1] += px * w4.y;
-                acc[k + 2] += px * w4.z;
-                acc[k + 3] += px * w4.w;
-              }
+            for (int j = 0; j < 2; ++j) {
+              uint32_t d = j ? d1 : d0;
+              pxp[2 * j] = des::pack_bf16((float)(d & 0xFF),
+                                          (float)((d >> 8) & 0xFF));
+              pxp[2 * j + 1] = des::pack_bf16((float)((d >> 16) & 0xFF),
+                                              (float)(d >> 24));
             }
+          } else {                                   // bf16 obs: already pairs
+            uint2 d0 = *reinterpret_cast<const uint2*>(row);
+            uint2 d1 = *reinterpret_cast<const uint2*>(row + 4);
+            pxp[0] = __builtin_bit_cast(des::bf16x2, d0.x);
+            pxp[1] = __builtin_bit_cast(des::bf16x2, d0.y);
+            pxp[2] = __builtin_bit_cast(des::bf16x2, d1.x);
+            pxp[3] = __builtin_bit_cast(des::bf16x2, d1.y);
+          }
+          const uint16_t* wrow = w_lds + ((ic * K1 + ky) * C1) * K1;
+#pragma unroll
+          for (int oc = 0; oc < C1; ++oc) {
+            uint4 wb = *reinterpret_cast<const uint4*>(wrow + oc * K1);
+            acc[oc] = des::dot2(pxp[0], __builtin_bit_cast(des::bf16x2, wb.x),
+                                acc[oc]);
+            acc[oc] = des::dot2(pxp[1], __builtin_bit_cast(des::bf16x2, wb.y),
+                                acc[oc]);
+            acc[oc] = des::dot2(pxp[2], __builtin_bit_cast(des::bf16x2, wb.z),
+                                acc[oc]);
+            acc[oc] = des::dot2(pxp[3], __builtin_bit_cast(des::bf16x2, wb.w),
+                                acc[oc]);
           }
         }
+      }
 #pragma unroll
-        for (int k = 0; k < 16; ++k) {
-          int oc = half * 16 + k;
-          float v = fmaxf(acc[k] + des::load1(b1 + oc), 0.f);
-          a1[oc * (O1 * O1) + pos] = des::f32_to_bf16(v);
-        }
+      for (int oc = 0; oc < C1; ++oc) {
+        float v = fmaxf(acc[oc] + des::load1(b1 + oc), 0.f);
+        a1[oc * (O1 * O1) + pos] = des::f32_to_bf16(v);
       }
     }
   }
@@ -156,16 +173,22 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
         if (pos >= O2 * O2) pos = O2 * O2 - 1;  // clamp; result discarded
         pbase[p] = (pos / O2) * S2 * O1 + (pos % O2) * S2;
       }
-      for (int icc = 0; icc < 512; icc += 4) {
-        float4 w4 = des::load4(w_lds + (icc >> 2) * (C2 * 4) + lane * 4);
-        const int4 off = *reinterpret_cast<const int4*>(t2 + icc);
+      // q = (ic, ky); the [q][oc][kx4] tile makes w a uint2 of 2 packed
+      // kx-pairs, and a1's kx run is contiguous (even index -> aligned b32
+      // pair reads) -> 2 LDS dwords + 2 dot2 per 4 MACs per position
+      for (int q = 0; q < 128; ++q) {
+        uint2 wp = *reinterpret_cast<const uint2*>(
+            w_lds + q * (C2 * 4) + lane * 4);
+        const int ab = t2[q << 2];        // ic*400 + ky*20 (kx = 0), even
 #pragma unroll
         for (int p = 0; p < 8; ++p) {
-          float a0 = des::bf16_to_f32(a1[pbase[p] + off.x]);
-          float a1v = des::bf16_to_f32(a1[pbase[p] + off.y]);
-          float a2v = des::bf16_to_f32(a1[pbase[p] + off.z]);
-          float a3v = des::bf16_to_f32(a1[pbase[p] + off.w]);
-          acc[p] += a0 * w4.x + a1v * w4.y + a2v * w4.z + a3v * w4.w;
+          const uint16_t* ap = a1 + pbase[p] + ab;
+          uint32_t a01 = *reinterpret_cast<const uint32_t*>(ap);
+          uint32_t a23 = *reinterpret_cast<const uint32_t*>(ap + 2);
+          acc[p] = des::dot2(__builtin_bit_cast(des::bf16x2, a01),
+                             __builtin_bit_cast(des::bf16x2, wp.x), acc[p]);
+          acc[p] = des::dot2(__builtin_bit_cast(des::bf16x2, a23),
+                             __builtin_bit_cast(des::bf16x2, wp.y), acc[p]);
         }
       }
 #pragma unroll
@@ -259,18 +282,22 @@ __global__ __launch_bounds__(256) void nature_fc_kernel(
     for (int ib = lane * 8; ib < FC_IN; ib += 64 * 8) {
       // x chunk (conflict-free b128: lane l -> banks 4l..4l+3 mod 64)
       uint4 xp = *reinterpret_cast<const uint4*>(xs + ib);
-      const uint16_t* xc = reinterpret_cast<const uint16_t*>(&xp);
       uint4 wa = *reinterpret_cast<const uint4*>(w0 + ib);
       uint4 wb = *reinterpret_cast<const uint4*>(w0 + FC_IN + ib);
       uint4 wc = *reinterpret_cast<const uint4*>(w0 + 2 * FC_IN + ib);
       uint4 wd = *reinterpret_cast<const uint4*>(w0 + 3 * FC_IN + ib);
+      const uint32_t* xq = reinterpret_cast<const uint32_t*>(&xp);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float xv = des::bf16_to_f32(xc[j]);
-        acc0 += xv * des::bf16_to_f32(reinterpret_cast<const uint16_t*>(&wa)[j]);
-        acc1 += xv * des::bf16_to_f32(reinterpret_cast<const uint16_t*>(&wb)[j]);
-        acc2 += xv * des::bf16_to_f32(reinterpret_cast<const uint16_t*>(&wc)[j]);
-        acc3 += xv * des::bf16_to_f32(reinterpret_cast<const uint16_t*>(&wd)[j]);
+      for (int j = 0; j < 4; ++j) {
+        des::bf16x2 xj = __builtin_bit_cast(des::bf16x2, xq[j]);
+        acc0 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+                 reinterpret_cast<const uint32_t*>(&wa)[j]), acc0);
+        acc1 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+                 reinterpret_cast<const uint32_t*>(&wb)[j]), acc1);
+        acc2 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+                 reinterpret_cast<const uint32_t*>(&wc)[j]), acc2);
+        acc3 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+                 reinterpret_cast<const uint32_t*>(&wd)[j]), acc3);
       }
     }
     acc0 = des::wave_reduce_sum_f(acc0);
