@@ -29,8 +29,9 @@ constexpr int FC_OUT = 512;
 constexpr int MAX_A = 32;
 
 // ---------------- kernel 1: convs ----------------
-// LDS carve (bytes, 16-aligned)
-constexpr int W_ELEMS = C2 * 512;                      // 32768 (conv2 tile)
+// LDS carve (bytes, 16-aligned).  conv2/conv3 weights staged in 32-oc
+// halves so the whole carve stays < 80 KiB -> 2 blocks/CU (latency hiding).
+constexpr int W_ELEMS = 32 * 576;                      // 18432 (conv3 half)
 constexpr int OFF_W = 0;                               // bf16 [W_ELEMS]
 constexpr int OFF_A1 = OFF_W + W_ELEMS * 2;            // bf16 [32*20*20]
 constexpr int OFF_A2 = OFF_A1 + C1 * O1 * O1 * 2;      // bf16 [64*9*9]
@@ -148,21 +149,23 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
   }
   __syncthreads();
 
-  // ---- stage W2: [oc][icc](K=512) -> [icc/4][oc][4] ---------------------
-  {
+  // ---- conv2 in two 32-oc halves: lane&31 = oc, lane>>5 = pos stream ----
+  // q = (ic, ky); the [q][oc][kx4] tile makes w a uint2 of 2 packed
+  // kx-pairs, and a1's kx run is contiguous (even index -> aligned b32
+  // pair reads) -> 2 LDS dwords + 2 dot2 per 4 MACs per position
+  for (int half = 0; half < 2; ++half) {
     const uint16_t* w2 = w + offs[2];
-    for (int e = tid; e < C2 * 512; e += 512) {
+    for (int e = tid; e < 32 * 512; e += 512) {
       int oc = e >> 9, icc = e & 511;
-      w_lds[(icc >> 2) * (C2 * 4) + oc * 4 + (icc & 3)] = w2[e];
+      w_lds[(icc >> 2) * 128 + oc * 4 + (icc & 3)] =
+          w2[(half * 32 + oc) * 512 + icc];
     }
-  }
-  __syncthreads();
-
-  // ---- conv2: lane = oc; 8 positions per wave iteration in registers ----
-  {
-    const uint16_t* b2 = w + offs[3];
-    const float bias = des::load1(b2 + lane);
-    for (int pb = wid * 8; pb < O2 * O2; pb += 64) {
+    __syncthreads();
+    {
+      const int oc = lane & 31;
+      const uint16_t* b2 = w + offs[3];
+      const float bias = des::load1(b2 + half * 32 + oc);
+      const int pb = (wid * 2 + (lane >> 5)) * 8;  // 16 streams x 8 = 128 >= 81
       float acc[8];
       int pbase[8];
 #pragma unroll
@@ -172,12 +175,8 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
         if (pos >= O2 * O2) pos = O2 * O2 - 1;  // clamp; result discarded
         pbase[p] = (pos / O2) * S2 * O1 + (pos % O2) * S2;
       }
-      // q = (ic, ky); the [q][oc][kx4] tile makes w a uint2 of 2 packed
-      // kx-pairs, and a1's kx run is contiguous (even index -> aligned b32
-      // pair reads) -> 2 LDS dwords + 2 dot2 per 4 MACs per position
       for (int q = 0; q < 128; ++q) {
-        uint2 wp = *reinterpret_cast<const uint2*>(
-            w_lds + q * (C2 * 4) + lane * 4);
+        uint2 wp = *reinterpret_cast<const uint2*>(w_lds + q * 128 + oc * 4);
         const int ab = t2[q << 2];        // ic*400 + ky*20 (kx = 0), even
 #pragma unroll
         for (int p = 0; p < 8; ++p) {
@@ -194,12 +193,12 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
       for (int p = 0; p < 8; ++p) {
         int pos = pb + p;
         if (pos < O2 * O2)
-          a2[lane * (O2 * O2) + pos] =
+          a2[(half * 32 + oc) * (O2 * O2) + pos] =
               des::f32_to_bf16(fmaxf(acc[p] + bias, 0.f));
       }
     }
+    __syncthreads();
   }
-  __syncthreads();
 
   // ---- conv3: two oc-halves (tile 36 KiB > W buffer? no: 32 oc = 18 KiB)
   {
