@@ -24,9 +24,8 @@ void des_launch_mlp_forward(const void*, bool, int64_t, const float*, int,
                             float*, int, int, int, const int64_t*, int,
                             hipStream_t);
 void des_launch_nature_cnn_forward(const uint16_t*, int64_t, const void*, bool,
-                                   const int*, uint16_t*, float*, int*, float*,
-                                   int, int, const int64_t*, const int64_t*,
-                                   hipStream_t);
+                                   const int*, int*, float*, int, int,
+                                   const int64_t*, hipStream_t);
 void des_launch_cartpole_reset(float*, int*, int64_t*, float*, uint8_t*,
                                float*, const int64_t*, int64_t, int,
                                hipStream_t);
@@ -156,30 +155,22 @@ void mlp_forward(torch::Tensor popw, torch::Tensor obs, torch::Tensor nmean,
 }
 
 void nature_cnn_forward(torch::Tensor popw, torch::Tensor frames,
-                        torch::Tensor slot_map, torch::Tensor fc_in,
-                        torch::Tensor fc_out, torch::Tensor actions,
-                        torch::Tensor logits, int64_t A, torch::Tensor offs,
-                        std::vector<int64_t> offs_host) {
+                        torch::Tensor slot_map, torch::Tensor actions,
+                        torch::Tensor logits, int64_t A, torch::Tensor offs) {
   CHECK_DEV(popw); CHECK_CONT(popw); CHECK_DEV(frames); CHECK_CONT(frames);
   CHECK_DEV(slot_map); CHECK_DEV(actions); CHECK_DEV(offs);
-  CHECK_DEV(fc_in); CHECK_CONT(fc_in); CHECK_DEV(fc_out); CHECK_CONT(fc_out);
   TORCH_CHECK(popw.dtype() == torch::kBFloat16, "pop weights must be bf16");
   TORCH_CHECK(offs.dtype() == torch::kInt64 && offs.numel() == 10);
-  TORCH_CHECK(offs_host.size() == 10);
   TORCH_CHECK(slot_map.dtype() == torch::kInt32 && slot_map.numel() == 4);
   TORCH_CHECK(actions.dtype() == torch::kInt32);
   int64_t P = popw.size(0);
-  TORCH_CHECK(fc_in.dtype() == torch::kBFloat16 && fc_in.numel() == P * 3136);
-  TORCH_CHECK(fc_out.dtype() == torch::kFloat32 && fc_out.numel() == P * 512);
   bool u8 = frames.dtype() == torch::kUInt8;
   TORCH_CHECK(u8 || frames.dtype() == torch::kBFloat16);
   float* lo = logits.numel() > 0 ? fptr(logits) : nullptr;
   des_launch_nature_cnn_forward(
       reinterpret_cast<const uint16_t*>(popw.data_ptr()), popw.size(1),
-      frames.data_ptr(), u8, slot_map.data_ptr<int>(),
-      reinterpret_cast<uint16_t*>(fc_in.data_ptr()), fptr(fc_out),
-      iptr(actions), lo, (int)A, (int)P, offs.data_ptr<int64_t>(),
-      offs_host.data(), cur_stream());
+      frames.data_ptr(), u8, slot_map.data_ptr<int>(), iptr(actions), lo,
+      (int)A, (int)P, offs.data_ptr<int64_t>(), cur_stream());
 }
 
 // ---------------- envs ----------------

@@ -2,19 +2,17 @@
 // K5: conv_forward_pop (Nature-CNN) — per-member policy forward as three
 // kernels sized for their regimes:
 //
-//   1. nature_conv_kernel   (block/member, 512 thr, ~105 KiB LDS)
-//      conv 8x8s4x32 -> 4x4s2x64 -> 3x3s1x64; activations in LDS; per-stage
-//      weight tiles staged coalesced into a conflict-free [k/4][oc][4] bf16
-//      layout; separable LDS offset tables kill per-MAC index math; conv1
-//      runs two 16-channel passes to stay under the VGPR cap (no spills).
-//      Writes the flattened conv output (fc input, [P,3136] bf16) to HBM.
-//   2. nature_fc_kernel     (grid P x 8 row-groups, 256 thr, ~7 KiB LDS)
-//      the HBM-bound stage (94% of weights): each wave streams 4 rows
-//      concurrently with bf16x8 (16 B) loads -> 4 independent load streams
-//      per wave x many blocks/CU saturate HBM (~3.2 MiB/member/step).
-//   3. nature_head_kernel   (block/member) 512->A + argmax (+ logits).
-//
-// Step floor at pop=8192/GPU: FC traffic 8192*3.2 MiB / 6.3 TBps ~= 4.3 ms.
+// ONE block-per-member kernel (512 thr, ~77 KiB LDS -> 2 blocks/CU):
+//   convs: 8x8s4x32 -> 4x4s2x64 -> 3x3s1x64 on packed-pair v_dot2c_f32_bf16,
+//     weight tiles staged in 32-oc halves to conflict-free [k/4][oc][4] bf16
+//     LDS layouts; activations live in LDS only;
+//   fc 3136->512: the HBM-bound stage (94% of weights) streamed with 4-row
+//     ILP bf16x8 (16 B) loads per wave;
+//   head 512->A + argmax fused.
+// Because 2 blocks co-reside per CU, one block's VALU-bound conv phase
+// overlaps the other block's HBM-bound FC stream: step time ~ max(conv, fc)
+// rather than the sum.  FC floor at pop=8192/GPU:
+// 8192 * 3.2 MiB / 6.3 TBps ~= 4.3 ms.
 #include "common.h"
 #include "gemv.h"
 
@@ -39,6 +37,12 @@ constexpr int OFF_A2 = OFF_A1 + C1 * O1 * O1 * 2;      // bf16 [64*9*9]
 constexpr int OFF_T2 = OFF_A2 + ((C2 * O2 * O2 * 2 + 15) & ~15);  // int [512]
 constexpr int OFF_T3 = OFF_T2 + 512 * 4;               // int [576]
 constexpr int CONV_LDS = OFF_T3 + 576 * 4;
+// fc input (3136 bf16), fc output (512 f32) and head logits overlay the a1
+// carve (dead after conv2): 6272 + 2048 + 128 B < 25600 B
+constexpr int OFF_A3 = OFF_A1;
+constexpr int OFF_FC = OFF_A3 + FC_IN * 2;
+constexpr int OFF_HD = OFF_FC + FC_OUT * 4;
+static_assert(OFF_HD + MAX_A * 4 <= OFF_A2, "fc overlay exceeds a1 carve");
 
 template <typename IT>
 __device__ __forceinline__ float obs_val(IT v);
@@ -54,7 +58,7 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
     const uint16_t* __restrict__ popw, int64_t row_stride,
     const IT* __restrict__ frames,     // [P][4][84][84]
     const int* __restrict__ slot_map,  // [4] logical->physical slot
-    uint16_t* __restrict__ fc_in,      // out: [P][3136] bf16
+    int* __restrict__ actions, float* __restrict__ logits_out, int A,
     int P, const int64_t* __restrict__ offs) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint16_t* w_lds = reinterpret_cast<uint16_t*>(smem + OFF_W);
@@ -62,6 +66,9 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
   uint16_t* a2 = reinterpret_cast<uint16_t*>(smem + OFF_A2);
   int* t2 = reinterpret_cast<int*>(smem + OFF_T2);
   int* t3 = reinterpret_cast<int*>(smem + OFF_T3);
+  uint16_t* a3 = reinterpret_cast<uint16_t*>(smem + OFF_A3);   // fc input
+  float* fcb = reinterpret_cast<float*>(smem + OFF_FC);
+  float* hd = reinterpret_cast<float*>(smem + OFF_HD);
 
   const int i = blockIdx.x;
   if (i >= P) return;
@@ -201,11 +208,10 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
     __syncthreads();
   }
 
-  // ---- conv3: two oc-halves (tile 36 KiB > W buffer? no: 32 oc = 18 KiB)
+  // ---- conv3: two oc-halves; output = fc input tile in LDS --------------
   {
     const uint16_t* w3 = w + offs[4];
     const uint16_t* b3 = w + offs[5];
-    uint16_t* out = fc_in + (int64_t)i * FC_IN;
     for (int half = 0; half < 2; ++half) {
       // stage [icc/4][oc32][4] for ocs [half*32, half*32+32)
       for (int e = tid; e < 32 * 576; e += 512) {
@@ -243,90 +249,59 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
         for (int p = 0; p < 4; ++p) {
           int pos = pb + p;
           if (pos < O3 * O3)
-            out[(half * 32 + oc) * (O3 * O3) + pos] =
+            a3[(half * 32 + oc) * (O3 * O3) + pos] =
                 des::f32_to_bf16(fmaxf(acc[p] + bias, 0.f));
         }
       }
       __syncthreads();
     }
   }
-}
 
-// ---------------- kernel 2: fc 3136 -> 512 (relu) ----------------
-// grid (P, 8): block handles 64 rows of one member; 4 waves x 4 rows
-// concurrently; 16-byte bf16x8 weight loads; x staged bf16 in LDS.
-__global__ __launch_bounds__(256) void nature_fc_kernel(
-    const uint16_t* __restrict__ popw, int64_t row_stride,
-    const uint16_t* __restrict__ fc_in,  // [P][3136]
-    float* __restrict__ fc_out,          // [P][512]
-    int P, int64_t wf_off, int64_t bf_off) {
-  __shared__ uint16_t xs[FC_IN];
-  const int i = blockIdx.x;
-  if (i >= P) return;
-  const int rg = blockIdx.y;           // row group: 64 rows
-  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
-  const uint16_t* x = fc_in + (int64_t)i * FC_IN;
-  for (int e = tid * 8; e < FC_IN; e += 256 * 8)
-    *reinterpret_cast<uint4*>(xs + e) =
-        *reinterpret_cast<const uint4*>(x + e);
-  __syncthreads();
-
-  const uint16_t* wf = popw + (int64_t)i * row_stride + wf_off;
-  const uint16_t* bf = popw + (int64_t)i * row_stride + bf_off;
-  // wave wid covers rows rg*64 + wid*16 + {0..15}, 4 at a time
-  for (int rr = 0; rr < 16; rr += 4) {
-    const int row0 = rg * 64 + wid * 16 + rr;
-    const uint16_t* w0 = wf + (int64_t)row0 * FC_IN;
-    float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
-    for (int ib = lane * 8; ib < FC_IN; ib += 64 * 8) {
-      // x chunk (conflict-free b128: lane l -> banks 4l..4l+3 mod 64)
-      uint4 xp = *reinterpret_cast<const uint4*>(xs + ib);
-      uint4 wa = *reinterpret_cast<const uint4*>(w0 + ib);
-      uint4 wb = *reinterpret_cast<const uint4*>(w0 + FC_IN + ib);
-      uint4 wc = *reinterpret_cast<const uint4*>(w0 + 2 * FC_IN + ib);
-      uint4 wd = *reinterpret_cast<const uint4*>(w0 + 3 * FC_IN + ib);
-      const uint32_t* xq = reinterpret_cast<const uint32_t*>(&xp);
+  // ---- fc 3136 -> 512 (relu): HBM-bound, 4 rows of ILP per wave ---------
+  {
+    const uint16_t* wf = w + offs[6];
+    const uint16_t* bf = w + offs[7];
+    for (int rr = 0; rr < 64; rr += 4) {
+      const int row0 = wid * 64 + rr;
+      const uint16_t* w0 = wf + (int64_t)row0 * FC_IN;
+      float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
+      for (int ib = lane * 8; ib < FC_IN; ib += 64 * 8) {
+        uint4 xp = *reinterpret_cast<const uint4*>(a3 + ib);
+        uint4 wa = *reinterpret_cast<const uint4*>(w0 + ib);
+        uint4 wb = *reinterpret_cast<const uint4*>(w0 + FC_IN + ib);
+        uint4 wc = *reinterpret_cast<const uint4*>(w0 + 2 * FC_IN + ib);
+        uint4 wd = *reinterpret_cast<const uint4*>(w0 + 3 * FC_IN + ib);
+        const uint32_t* xq = reinterpret_cast<const uint32_t*>(&xp);
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        des::bf16x2 xj = __builtin_bit_cast(des::bf16x2, xq[j]);
-        acc0 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
-                 reinterpret_cast<const uint32_t*>(&wa)[j]), acc0);
-        acc1 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
-                 reinterpret_cast<const uint32_t*>(&wb)[j]), acc1);
-        acc2 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
-                 reinterpret_cast<const uint32_t*>(&wc)[j]), acc2);
-        acc3 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
-                 reinterpret_cast<const uint32_t*>(&wd)[j]), acc3);
+        for (int j = 0; j < 4; ++j) {
+          des::bf16x2 xj = __builtin_bit_cast(des::bf16x2, xq[j]);
+          acc0 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+                   reinterpret_cast<const uint32_t*>(&wa)[j]), acc0);
+          acc1 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+                   reinterpret_cast<const uint32_t*>(&wb)[j]), acc1);
+          acc2 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+                   reinterpret_cast<const uint32_t*>(&wc)[j]), acc2);
+          acc3 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+                   reinterpret_cast<const uint32_t*>(&wd)[j]), acc3);
+        }
+      }
+      acc0 = des::wave_reduce_sum_f(acc0);
+      acc1 = des::wave_reduce_sum_f(acc1);
+      acc2 = des::wave_reduce_sum_f(acc2);
+      acc3 = des::wave_reduce_sum_f(acc3);
+      if (lane < 4) {
+        float acc = lane == 0 ? acc0 : lane == 1 ? acc1
+                    : lane == 2 ? acc2 : acc3;
+        float bias = des::bf16_to_f32(bf[row0 + lane]);
+        fcb[row0 + lane] = fmaxf(acc + bias, 0.f);
       }
     }
-    acc0 = des::wave_reduce_sum_f(acc0);
-    acc1 = des::wave_reduce_sum_f(acc1);
-    acc2 = des::wave_reduce_sum_f(acc2);
-    acc3 = des::wave_reduce_sum_f(acc3);
-    if (lane < 4) {
-      float acc = lane == 0 ? acc0 : lane == 1 ? acc1 : lane == 2 ? acc2 : acc3;
-      float bias = des::bf16_to_f32(bf[row0 + lane]);
-      fc_out[(int64_t)i * FC_OUT + row0 + lane] = fmaxf(acc + bias, 0.f);
-    }
   }
-}
-
-// ---------------- kernel 3: head 512 -> A + argmax ----------------
-__global__ __launch_bounds__(256) void nature_head_kernel(
-    const uint16_t* __restrict__ popw, int64_t row_stride,
-    const float* __restrict__ fc_out, int* __restrict__ actions,
-    float* __restrict__ logits_out, int A, int P, int64_t wh_off,
-    int64_t bh_off) {
-  __shared__ float xs[FC_OUT];
-  __shared__ float hd[MAX_A];
-  const int i = blockIdx.x;
-  if (i >= P) return;
-  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
-  for (int e = tid; e < FC_OUT; e += 256) xs[e] = fc_out[(int64_t)i * FC_OUT + e];
   __syncthreads();
-  const uint16_t* w = popw + (int64_t)i * row_stride;
-  des::gemv_rows<uint16_t, 0>(w + wh_off, w + bh_off, xs, hd, FC_OUT, A,
-                              FC_OUT, lane, wid, 4);
+
+  // ---- head 512 -> A + argmax ------------------------------------------
+  des::gemv_rows<uint16_t, 0>(w + offs[8], w + offs[9], fcb, hd, FC_OUT, A,
+                              FC_OUT, lane, wid, 8);
   __syncthreads();
   if (logits_out && tid < A) logits_out[(int64_t)i * A + tid] = hd[tid];
   if (tid == 0) {
@@ -342,9 +317,8 @@ __global__ __launch_bounds__(256) void nature_head_kernel(
 
 extern "C" void des_launch_nature_cnn_forward(
     const uint16_t* popw, int64_t row_stride, const void* frames, bool obs_u8,
-    const int* slot_map_dev, uint16_t* fc_in, float* fc_out, int* actions,
-    float* logits_out, int A, int P, const int64_t* offs_dev,
-    const int64_t* offs_host, hipStream_t stream) {
+    const int* slot_map_dev, int* actions, float* logits_out, int A, int P,
+    const int64_t* offs_dev, hipStream_t stream) {
   static bool attr_set = false;
   if (!attr_set) {
     hipFuncSetAttribute((const void*)nature_conv_kernel<uint8_t>,
@@ -356,17 +330,11 @@ extern "C" void des_launch_nature_cnn_forward(
   if (obs_u8)
     hipLaunchKernelGGL(nature_conv_kernel<uint8_t>, dim3((uint32_t)P),
                        dim3(512), CONV_LDS, stream, popw, row_stride,
-                       (const uint8_t*)frames, slot_map_dev, fc_in, P,
-                       offs_dev);
+                       (const uint8_t*)frames, slot_map_dev, actions,
+                       logits_out, A, P, offs_dev);
   else
     hipLaunchKernelGGL(nature_conv_kernel<uint16_t>, dim3((uint32_t)P),
                        dim3(512), CONV_LDS, stream, popw, row_stride,
-                       (const uint16_t*)frames, slot_map_dev, fc_in, P,
-                       offs_dev);
-  hipLaunchKernelGGL(nature_fc_kernel, dim3((uint32_t)P, 8), dim3(256), 0,
-                     stream, popw, row_stride, fc_in, fc_out, P, offs_host[6],
-                     offs_host[7]);
-  hipLaunchKernelGGL(nature_head_kernel, dim3((uint32_t)P), dim3(256), 0,
-                     stream, popw, row_stride, fc_out, actions, logits_out, A,
-                     P, offs_host[8], offs_host[9]);
+                       (const uint16_t*)frames, slot_map_dev, actions,
+                       logits_out, A, P, offs_dev);
 }

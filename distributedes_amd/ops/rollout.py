@@ -81,9 +81,6 @@ class HipRollout:
         self.P = P
         z = lambda *s, dt=torch.float32: torch.zeros(*s, dtype=dt, device=dev)  # noqa: E731
         self._empty = torch.empty(0, device=dev)
-        if self.kind == "nature_cnn":
-            self.fc_in = z(P, 3136, dt=torch.bfloat16)
-            self.fc_out = z(P, 512)
         self.fitness = z(P)
         self.alive = z(P, dt=torch.uint8)
         self.ep_len = z(P)
@@ -167,10 +164,9 @@ class HipRollout:
             else:
                 frames = self.obs_bf16
                 sm = self.slot_map
-            self.ext.nature_cnn_forward(pop_w, frames, sm, self.fc_in,
-                                        self.fc_out, self.actions_i,
+            self.ext.nature_cnn_forward(pop_w, frames, sm, self.actions_i,
                                         self._empty, self.env.num_actions,
-                                        self.offs_t, self.offs_h)
+                                        self.offs_t)
 
     def _env_step(self, slot: int):
         e = self.env_name

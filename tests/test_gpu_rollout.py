@@ -89,12 +89,8 @@ def test_nature_cnn_forward_logits_match_ref(ext):
     slot_map = torch.tensor([0, 1, 2, 3], dtype=torch.int32).cuda()
     acts = torch.zeros(P, dtype=torch.int32).cuda()
     logits = torch.zeros(P, 18).cuda()
-    offs_h = nature_cnn_offsets(pol.spec)
-    offs = torch.tensor(offs_h, dtype=torch.int64).cuda()
-    fc_in = torch.zeros(P, 3136, dtype=torch.bfloat16).cuda()
-    fc_out = torch.zeros(P, 512).cuda()
-    ext.nature_cnn_forward(popw, obs_u8, slot_map, fc_in, fc_out, acts,
-                           logits, 18, offs, offs_h)
+    offs = torch.tensor(nature_cnn_offsets(pol.spec), dtype=torch.int64).cuda()
+    ext.nature_cnn_forward(popw, obs_u8, slot_map, acts, logits, 18, offs)
     # reference: fp32 math on the same bf16-rounded weights
     ref = _nature_ref_logits(pol, flat.cpu().float(), obs_u8.cpu().float())
     # kernel stores intermediate activations in bf16 -> tolerate ~1% rel
@@ -134,10 +130,7 @@ def test_nature_cnn_slot_map_rolls_frames(ext):
     flat = (torch.randn(P, pol.dim) * 0.05).to(torch.bfloat16).cuda()
     popw = _pad(flat)
     frames = (torch.rand(P, 4, 84, 84) < 0.2).to(torch.uint8).cuda()
-    offs_h = nature_cnn_offsets(pol.spec)
-    offs = torch.tensor(offs_h, dtype=torch.int64).cuda()
-    fc_in = torch.zeros(P, 3136, dtype=torch.bfloat16).cuda()
-    fc_out = torch.zeros(P, 512).cuda()
+    offs = torch.tensor(nature_cnn_offsets(pol.spec), dtype=torch.int64).cuda()
     outs = []
     for r in range(4):
         # rolled[:, (c + r) % 4] == frames[:, c]
@@ -146,8 +139,7 @@ def test_nature_cnn_slot_map_rolls_frames(ext):
                           dtype=torch.int32).cuda()
         logits = torch.zeros(P, 18).cuda()
         acts = torch.zeros(P, dtype=torch.int32).cuda()
-        ext.nature_cnn_forward(popw, rolled, sm, fc_in, fc_out, acts, logits,
-                               18, offs, offs_h)
+        ext.nature_cnn_forward(popw, rolled, sm, acts, logits, 18, offs)
         outs.append(logits.cpu())
     for r in (1, 2, 3):
         torch.testing.assert_close(outs[0], outs[r])
