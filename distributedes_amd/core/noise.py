@@ -131,14 +131,32 @@ def member_offset(run_seed: int, generation: int, pair_index: int,
     return int((s % ((span // 16) + 1)) * 16) if span >= 16 else int(s % span)
 
 
+def _splitmix64_np(x: np.ndarray) -> np.ndarray:
+    """Vectorized splitmix64 (uint64 wraparound == the scalar version)."""
+    with np.errstate(over="ignore"):
+        x = (x + np.uint64(0x9E3779B97F4A7C15))
+        z = x
+        z = (z ^ (z >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+        z = (z ^ (z >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+        return z ^ (z >> np.uint64(31))
+
+
 def member_offsets(run_seed: int, generation: int, pop: int, antithetic: bool,
                    table_size: int, dim: int) -> tuple:
     """(offsets[pop] int64, signs[pop] float32) for the whole global population."""
-    offs = np.empty(pop, dtype=np.int64)
+    with np.errstate(over="ignore"):
+        pairs = (np.arange(pop, dtype=np.uint64) // np.uint64(2)
+                 if antithetic else np.arange(pop, dtype=np.uint64))
+        base = np.uint64(splitmix64((run_seed ^ (generation * 0x9E3779B97F4A7C15)) & _U64))
+        seeds = _splitmix64_np(np.full(pop, base, dtype=np.uint64)
+                               ^ (pairs * np.uint64(0xBF58476D1CE4E5B9)))
+    span = table_size - dim
+    assert span > 0, "noise table smaller than parameter dim"
+    if span >= 16:
+        offs = ((seeds % np.uint64(span // 16 + 1)) * np.uint64(16)).astype(np.int64)
+    else:
+        offs = (seeds % np.uint64(span)).astype(np.int64)
     signs = np.ones(pop, dtype=np.float32)
-    for i in range(pop):
-        pair = i // 2 if antithetic else i
-        offs[i] = member_offset(run_seed, generation, pair, table_size, dim)
-        if antithetic and (i & 1):
-            signs[i] = -1.0
+    if antithetic:
+        signs[1::2] = -1.0
     return offs, signs

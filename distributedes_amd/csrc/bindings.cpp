@@ -51,6 +51,14 @@ void des_launch_humanoid_step(float*, int*, const float*, float*, uint8_t*,
                               float*, int, int, hipStream_t);
 void des_launch_normalizer_observe(const float*, double*, double*, int, int,
                                    hipStream_t);
+void des_launch_impala_conv3(const uint16_t*, uint16_t*, const uint16_t*,
+                             const uint16_t*, int64_t, int64_t, int64_t, int,
+                             bool, bool, int, hipStream_t);
+void des_launch_impala_pool(const uint16_t*, uint16_t*, int, int, int,
+                            hipStream_t);
+void des_launch_impala_fc_head(const uint16_t*, int64_t, const uint16_t*,
+                               int*, float*, int, int, int64_t, int64_t,
+                               int64_t, int64_t, hipStream_t);
 }
 
 namespace {
@@ -265,6 +273,42 @@ void humanoid_step(torch::Tensor s, torch::Tensor steps, torch::Tensor actions,
                            cur_stream());
 }
 
+const uint16_t* u16p(const torch::Tensor& t) {
+  return reinterpret_cast<const uint16_t*>(t.data_ptr());
+}
+
+void impala_conv(torch::Tensor in, torch::Tensor out, torch::Tensor skip,
+                 torch::Tensor popw, int64_t w_off, int64_t b_off,
+                 int64_t shape, bool relu_in, bool add_skip) {
+  CHECK_DEV(in); CHECK_CONT(in); CHECK_DEV(out); CHECK_CONT(out);
+  CHECK_DEV(popw); CHECK_CONT(popw);
+  TORCH_CHECK(in.dtype() == torch::kBFloat16 && out.dtype() == torch::kBFloat16);
+  int P = popw.size(0);
+  const uint16_t* sk = add_skip ? u16p(skip) : nullptr;
+  des_launch_impala_conv3(u16p(in), (uint16_t*)out.data_ptr(), sk, u16p(popw),
+                          popw.size(1), w_off, b_off, (int)shape, relu_in,
+                          add_skip, P, cur_stream());
+}
+
+void impala_pool(torch::Tensor in, torch::Tensor out, int64_t C, int64_t HW) {
+  CHECK_DEV(in); CHECK_CONT(in); CHECK_DEV(out); CHECK_CONT(out);
+  int P = in.size(0);
+  des_launch_impala_pool(u16p(in), (uint16_t*)out.data_ptr(), (int)C, (int)HW,
+                         P, cur_stream());
+}
+
+void impala_fc_head(torch::Tensor popw, torch::Tensor act,
+                    torch::Tensor actions, torch::Tensor logits, int64_t A,
+                    int64_t wf, int64_t bf, int64_t wh, int64_t bh) {
+  CHECK_DEV(popw); CHECK_CONT(popw); CHECK_DEV(act); CHECK_CONT(act);
+  CHECK_DEV(actions);
+  int P = popw.size(0);
+  float* lo = logits.numel() > 0 ? fptr(logits) : nullptr;
+  des_launch_impala_fc_head(u16p(popw), popw.size(1), u16p(act),
+                            iptr(actions), lo, (int)A, P, wf, bf, wh, bh,
+                            cur_stream());
+}
+
 void normalizer_observe(torch::Tensor obs, torch::Tensor sum,
                         torch::Tensor sumsq) {
   CHECK_DEV(obs); CHECK_CONT(obs);
@@ -294,4 +338,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("humanoid_reset", &humanoid_reset);
   m.def("humanoid_step", &humanoid_step);
   m.def("normalizer_observe", &normalizer_observe);
+  m.def("impala_conv", &impala_conv, "K6: 3x3 pad-1 conv (pop, tiled)");
+  m.def("impala_pool", &impala_pool, "K6: 3x3 s2 pad-1 max pool");
+  m.def("impala_fc_head", &impala_fc_head, "K6: fc 2048->256 + head + argmax");
 }

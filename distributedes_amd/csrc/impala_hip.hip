@@ -1,0 +1,250 @@
+#include "hip/hip_runtime.h"
+// K6: conv_forward_pop (IMPALA-CNN) — per-member forward for the Procgen-dim
+// config (BASELINE config #5).
+//
+// Unlike the Nature-CNN (one fused block/member kernel), IMPALA's stage-1
+// activations (64x64x16 bf16 = 128 KiB) exceed LDS, so the net runs as a
+// sequence of generic batched kernels over [P, C, H, W] bf16 scratch:
+//   impala_conv3: 3x3 pad-1 conv, spatial 16x16 tiles, halo-staged input +
+//     [k][co] transposed weight tile in LDS (~40 KiB -> 4 blocks/CU);
+//     optional fused input-relu (pre-activation residual) and skip-add.
+//   impala_pool:  3x3 stride-2 pad-1 max pool.
+//   impala_fc_head: relu-flatten 2048 -> fc 256 (4-row-ILP bf16x8 dot2
+//     stream) -> head A + argmax.
+#include "common.h"
+#include "gemv.h"
+
+namespace {
+
+template <int CI, int CO, int HW, int TS, bool RELU_IN, bool ADD_SKIP>
+__global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
+    const uint16_t* __restrict__ in,    // [P][CI][HW][HW]
+    uint16_t* __restrict__ out,         // [P][CO][HW][HW]
+    const uint16_t* __restrict__ skip,  // nullable [P][CO][HW][HW]
+    const uint16_t* __restrict__ popw, int64_t row_stride, int64_t w_off,
+    int64_t b_off, int P) {
+  constexpr int TH = TS + 2;
+  __shared__ uint16_t a_lds[CI * TH * TH];
+  __shared__ uint16_t w_lds[CI * 9 * CO];
+  const int i = blockIdx.x;
+  if (i >= P) return;
+  constexpr int NT = HW / TS;
+  const int ty0 = (blockIdx.y / NT) * TS, tx0 = (blockIdx.y % NT) * TS;
+  const int tid = threadIdx.x;
+  const uint16_t* w = popw + (int64_t)i * row_stride;
+
+  // stage weights transposed [icc][co]
+  const uint16_t* wsrc = w + w_off;
+  for (int e = tid; e < CO * CI * 9; e += TS * TS) {
+    int co = e / (CI * 9), icc = e % (CI * 9);
+    w_lds[icc * CO + co] = wsrc[e];
+  }
+  // stage input halo tile (zero-padded at image edges)
+  const uint16_t* src = in + (int64_t)i * CI * HW * HW;
+  for (int e = tid; e < CI * TH * TH; e += TS * TS) {
+    int ci = e / (TH * TH), r = e % (TH * TH);
+    int y = ty0 + r / TH - 1, x = tx0 + r % TH - 1;
+    uint16_t v = 0;
+    if (y >= 0 && y < HW && x >= 0 && x < HW) {
+      v = src[(ci * HW + y) * HW + x];
+      if (RELU_IN && (v & 0x8000)) v = 0;  // relu on bf16: clear negatives
+    } else if (RELU_IN) {
+      v = 0;
+    }
+    a_lds[e] = v;
+  }
+  __syncthreads();
+
+  const int py = tid / TS, px = tid % TS;          // position in tile
+  float acc[CO];
+#pragma unroll
+  for (int c = 0; c < CO; ++c) acc[c] = 0.f;
+  for (int ci = 0; ci < CI; ++ci) {
+#pragma unroll
+    for (int dy = 0; dy < 3; ++dy) {
+#pragma unroll
+      for (int dx = 0; dx < 3; ++dx) {
+        float a = des::bf16_to_f32(
+            a_lds[(ci * TH + py + dy) * TH + px + dx]);
+        const uint16_t* wr = w_lds + ((ci * 9) + dy * 3 + dx) * CO;
+#pragma unroll
+        for (int c = 0; c < CO; c += 4) {
+          float4 w4 = des::load4(wr + c);
+          acc[c] += a * w4.x;
+          acc[c + 1] += a * w4.y;
+          acc[c + 2] += a * w4.z;
+          acc[c + 3] += a * w4.w;
+        }
+      }
+    }
+  }
+  const uint16_t* bias = w + b_off;
+  uint16_t* dst = out + (int64_t)i * CO * HW * HW;
+  const uint16_t* sk =
+      ADD_SKIP ? skip + (int64_t)i * CO * HW * HW : nullptr;
+  const int oy = ty0 + py, ox = tx0 + px;
+#pragma unroll
+  for (int c = 0; c < CO; ++c) {
+    float v = acc[c] + des::load1(bias + c);
+    if (ADD_SKIP) v += des::bf16_to_f32(sk[(c * HW + oy) * HW + ox]);
+    dst[(c * HW + oy) * HW + ox] = des::f32_to_bf16(v);
+  }
+}
+
+template <int C, int HW>  // output is HW/2 x HW/2 (3x3 s2 pad1)
+__global__ void impala_pool_kernel(const uint16_t* __restrict__ in,
+                                   uint16_t* __restrict__ out, int P) {
+  constexpr int HO = HW / 2;
+  int64_t n = (int64_t)P * C * HO * HO;
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= n) return;
+  int ox = idx % HO, oy = (idx / HO) % HO;
+  int64_t pc = idx / (HO * HO);  // member*C + c
+  const uint16_t* src = in + pc * HW * HW;
+  float m = -3.0e38f;
+  for (int dy = -1; dy <= 1; ++dy) {
+    int y = 2 * oy + dy;
+    if (y < 0 || y >= HW) continue;
+    for (int dx = -1; dx <= 1; ++dx) {
+      int x = 2 * ox + dx;
+      if (x < 0 || x >= HW) continue;
+      m = fmaxf(m, des::bf16_to_f32(src[y * HW + x]));
+    }
+  }
+  out[idx] = des::f32_to_bf16(m);
+}
+
+// relu-flatten 2048 -> fc 256 relu -> head A + argmax
+constexpr int IFC_IN = 2048, IFC_OUT = 256, IMAX_A = 32;
+
+__global__ __launch_bounds__(256) void impala_fc_head_kernel(
+    const uint16_t* __restrict__ popw, int64_t row_stride,
+    const uint16_t* __restrict__ act,  // [P][2048] (pre-relu)
+    int* __restrict__ actions, float* __restrict__ logits_out, int A, int P,
+    int64_t wf, int64_t bf, int64_t wh, int64_t bh) {
+  __shared__ uint16_t xs[IFC_IN];
+  __shared__ float fcb[IFC_OUT];
+  __shared__ float hd[IMAX_A];
+  const int i = blockIdx.x;
+  if (i >= P) return;
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  const uint16_t* src = act + (int64_t)i * IFC_IN;
+  for (int e = tid; e < IFC_IN; e += 256) {
+    uint16_t v = src[e];
+    xs[e] = (v & 0x8000) ? 0 : v;  // relu
+  }
+  __syncthreads();
+  const uint16_t* w = popw + (int64_t)i * row_stride;
+  const uint16_t* w0base = w + wf;
+  const uint16_t* bfp = w + bf;
+  for (int rr = 0; rr < 64; rr += 4) {
+    const int row0 = wid * 64 + rr;
+    const uint16_t* w0 = w0base + (int64_t)row0 * IFC_IN;
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {       // 2048 = 4 * 512 exactly
+      int ib = lane * 8 + it * 512;
+      uint4 xp = *reinterpret_cast<const uint4*>(xs + ib);
+      uint4 wa = *reinterpret_cast<const uint4*>(w0 + ib);
+      uint4 wb = *reinterpret_cast<const uint4*>(w0 + IFC_IN + ib);
+      uint4 wc = *reinterpret_cast<const uint4*>(w0 + 2 * IFC_IN + ib);
+      uint4 wd = *reinterpret_cast<const uint4*>(w0 + 3 * IFC_IN + ib);
+      const uint32_t* xq = reinterpret_cast<const uint32_t*>(&xp);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        des::bf16x2 xj = __builtin_bit_cast(des::bf16x2, xq[j]);
+        a0 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+               reinterpret_cast<const uint32_t*>(&wa)[j]), a0);
+        a1 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+               reinterpret_cast<const uint32_t*>(&wb)[j]), a1);
+        a2 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+               reinterpret_cast<const uint32_t*>(&wc)[j]), a2);
+        a3 = des::dot2(xj, __builtin_bit_cast(des::bf16x2,
+               reinterpret_cast<const uint32_t*>(&wd)[j]), a3);
+      }
+    }
+    a0 = des::wave_reduce_sum_f(a0);
+    a1 = des::wave_reduce_sum_f(a1);
+    a2 = des::wave_reduce_sum_f(a2);
+    a3 = des::wave_reduce_sum_f(a3);
+    if (lane < 4) {
+      float acc = lane == 0 ? a0 : lane == 1 ? a1 : lane == 2 ? a2 : a3;
+      fcb[row0 + lane] =
+          fmaxf(acc + des::bf16_to_f32(bfp[row0 + lane]), 0.f);
+    }
+  }
+  __syncthreads();
+  des::gemv_rows<uint16_t, 0>(w + wh, w + bh, fcb, hd, IFC_OUT, A, IFC_OUT,
+                              lane, wid, 4);
+  __syncthreads();
+  if (logits_out && tid < A) logits_out[(int64_t)i * A + tid] = hd[tid];
+  if (tid == 0) {
+    int best = 0;
+    float bv = hd[0];
+    for (int a = 1; a < A; ++a)
+      if (hd[a] > bv) { bv = hd[a]; best = a; }
+    actions[i] = best;
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+// shape key: 0:(3,16,64) 1:(16,16,32) 2:(16,32,32) 3:(32,32,16) 4:(32,32,8)
+void des_launch_impala_conv3(const uint16_t* in, uint16_t* out,
+                             const uint16_t* skip, const uint16_t* popw,
+                             int64_t row_stride, int64_t w_off, int64_t b_off,
+                             int shape, bool relu_in, bool add_skip, int P,
+                             hipStream_t stream) {
+#define CASE(S, CI, CO, HW, TS)                                               \
+  if (shape == S) {                                                           \
+    dim3 grid((uint32_t)P, (HW / TS) * (HW / TS));                            \
+    if (relu_in && add_skip)                                                  \
+      hipLaunchKernelGGL((impala_conv3_kernel<CI, CO, HW, TS, true, true>),   \
+                         grid, dim3(TS * TS), 0, stream, in, out, skip, popw, \
+                         row_stride, w_off, b_off, P);                        \
+    else if (relu_in)                                                         \
+      hipLaunchKernelGGL((impala_conv3_kernel<CI, CO, HW, TS, true, false>),  \
+                         grid, dim3(TS * TS), 0, stream, in, out, skip, popw, \
+                         row_stride, w_off, b_off, P);                        \
+    else                                                                      \
+      hipLaunchKernelGGL((impala_conv3_kernel<CI, CO, HW, TS, false, false>), \
+                         grid, dim3(TS * TS), 0, stream, in, out, skip, popw, \
+                         row_stride, w_off, b_off, P);                        \
+    return;                                                                   \
+  }
+  CASE(0, 3, 16, 64, 16)
+  CASE(1, 16, 16, 32, 16)
+  CASE(2, 16, 32, 32, 16)
+  CASE(3, 32, 32, 16, 16)
+  CASE(4, 32, 32, 8, 8)
+#undef CASE
+}
+
+void des_launch_impala_pool(const uint16_t* in, uint16_t* out, int C, int HW,
+                            int P, hipStream_t stream) {
+  int64_t n = (int64_t)P * C * (HW / 2) * (HW / 2);
+  int g = (int)((n + 255) / 256);
+  if (C == 16 && HW == 64)
+    hipLaunchKernelGGL((impala_pool_kernel<16, 64>), dim3(g), dim3(256), 0,
+                       stream, in, out, P);
+  else if (C == 32 && HW == 32)
+    hipLaunchKernelGGL((impala_pool_kernel<32, 32>), dim3(g), dim3(256), 0,
+                       stream, in, out, P);
+  else if (C == 32 && HW == 16)
+    hipLaunchKernelGGL((impala_pool_kernel<32, 16>), dim3(g), dim3(256), 0,
+                       stream, in, out, P);
+}
+
+void des_launch_impala_fc_head(const uint16_t* popw, int64_t row_stride,
+                               const uint16_t* act, int* actions,
+                               float* logits_out, int A, int P, int64_t wf,
+                               int64_t bf, int64_t wh, int64_t bh,
+                               hipStream_t stream) {
+  hipLaunchKernelGGL(impala_fc_head_kernel, dim3((uint32_t)P), dim3(256), 0,
+                     stream, popw, row_stride, act, actions, logits_out, A, P,
+                     wf, bf, wh, bh);
+}
+
+}  // extern "C"

@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import torch
 
-from ..models import MLPPolicy, NatureCNNPolicy
+from ..models import ImpalaCNNPolicy, MLPPolicy, NatureCNNPolicy
 from ..models.base import ParamSpec
 
 
@@ -37,6 +37,22 @@ def nature_cnn_offsets(spec: ParamSpec):
     offs = [spec._by_name[n].offset for n in names]
     assert all(o % 4 == 0 for o in offs)
     return offs
+
+
+def impala_offsets(spec: ParamSpec):
+    o = lambda n: spec._by_name[n].offset  # noqa: E731
+    d = {}
+    for sg in (1, 2, 3):
+        d[f"s{sg}"] = (o(f"stage{sg}_conv.weight"), o(f"stage{sg}_conv.bias"))
+        for r in (1, 2):
+            d[f"s{sg}r{r}c1"] = (o(f"stage{sg}_res{r}_conv1.weight"),
+                                 o(f"stage{sg}_res{r}_conv1.bias"))
+            d[f"s{sg}r{r}c2"] = (o(f"stage{sg}_res{r}_conv2.weight"),
+                                 o(f"stage{sg}_res{r}_conv2.bias"))
+    d["fc"] = (o("fc.weight"), o("fc.bias"))
+    d["head"] = (o("head.weight"), o("head.bias"))
+    assert all(v % 4 == 0 for pair in d.values() for v in pair)
+    return d
 
 
 class HipRollout:
@@ -64,6 +80,9 @@ class HipRollout:
             self.offs_h = nature_cnn_offsets(model.spec)
             self.offs_t = torch.tensor(self.offs_h, dtype=torch.int64,
                                        device=device)
+        elif isinstance(model, ImpalaCNNPolicy):
+            self.kind = "impala"
+            self.ioffs = impala_offsets(model.spec)
         else:
             raise NotImplementedError(type(model))
 
@@ -71,8 +90,8 @@ class HipRollout:
     def supports(cls, cfg, model) -> bool:
         if cfg.env not in cls.SUPPORTED_ENVS:
             return False
-        if isinstance(model, NatureCNNPolicy):
-            return cfg.precision == "bf16"  # kernel consumes bf16 pop weights
+        if isinstance(model, (NatureCNNPolicy, ImpalaCNNPolicy)):
+            return cfg.precision == "bf16"  # kernels consume bf16 pop weights
         return isinstance(model, MLPPolicy) and len(model.hidden_sizes) <= 2
 
     # ------------------------------------------------------------------ #
@@ -110,6 +129,17 @@ class HipRollout:
             self.obs_bf16 = z(P, c, hw, hw, dt=torch.bfloat16)
             self.slot_map = torch.tensor([0, 1, 2, 3], dtype=torch.int32,
                                          device=dev)
+        if self.kind == "impala":
+            bf = torch.bfloat16
+            self.ib64 = z(P, 16, 64, 64, dt=bf)
+            self.ix32 = z(P, 16, 32, 32, dt=bf)
+            self.it32 = z(P, 16, 32, 32, dt=bf)
+            self.ib32 = z(P, 32, 32, 32, dt=bf)
+            self.ix16 = z(P, 32, 16, 16, dt=bf)
+            self.it16 = z(P, 32, 16, 16, dt=bf)
+            self.ib16 = z(P, 32, 16, 16, dt=bf)
+            self.ix8 = z(P, 32, 8, 8, dt=bf)
+            self.it8 = z(P, 32, 8, 8, dt=bf)
         elif e == "humanoid_synth":
             self.hu_s = z(P, 376)
             self.hu_steps = z(P, dt=torch.int32)
@@ -157,7 +187,7 @@ class HipRollout:
             self.ext.mlp_forward(pop_w, obs, nmean, ninv, clip, act,
                                  self._empty, self.h1, self.h2,
                                  self.env.num_actions, self.offs)
-        else:  # nature_cnn
+        elif self.kind == "nature_cnn":
             if self.env_name == "pong":
                 frames = self.frames
                 sm = self.slot_maps[(slot_newest + 1) % 4]
@@ -167,6 +197,36 @@ class HipRollout:
             self.ext.nature_cnn_forward(pop_w, frames, sm, self.actions_i,
                                         self._empty, self.env.num_actions,
                                         self.offs_t)
+        else:  # impala
+            self._impala_forward(pop_w)
+
+    def _impala_forward(self, pop_w, logits=None):
+        ext, io, E = self.ext, self.ioffs, self._empty
+        conv = lambda i, o, sk, name, shape, relu, add: ext.impala_conv(  # noqa: E731
+            i, o, sk if sk is not None else E, pop_w, io[name][0], io[name][1],
+            shape, relu, add)
+        # stage 1
+        conv(self.obs_bf16, self.ib64, None, "s1", 0, False, False)
+        ext.impala_pool(self.ib64, self.ix32, 16, 64)
+        for r in (1, 2):
+            conv(self.ix32, self.it32, None, f"s1r{r}c1", 1, True, False)
+            conv(self.it32, self.ix32, self.ix32, f"s1r{r}c2", 1, True, True)
+        # stage 2
+        conv(self.ix32, self.ib32, None, "s2", 2, False, False)
+        ext.impala_pool(self.ib32, self.ix16, 32, 32)
+        for r in (1, 2):
+            conv(self.ix16, self.it16, None, f"s2r{r}c1", 3, True, False)
+            conv(self.it16, self.ix16, self.ix16, f"s2r{r}c2", 3, True, True)
+        # stage 3
+        conv(self.ix16, self.ib16, None, "s3", 3, False, False)
+        ext.impala_pool(self.ib16, self.ix8, 32, 16)
+        for r in (1, 2):
+            conv(self.ix8, self.it8, None, f"s3r{r}c1", 4, True, False)
+            conv(self.it8, self.ix8, self.ix8, f"s3r{r}c2", 4, True, True)
+        ext.impala_fc_head(pop_w, self.ix8.view(self.P, -1), self.actions_i,
+                           logits if logits is not None else E,
+                           self.env.num_actions, io["fc"][0], io["fc"][1],
+                           io["head"][0], io["head"][1])
 
     def _env_step(self, slot: int):
         e = self.env_name
