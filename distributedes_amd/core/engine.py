@@ -181,6 +181,7 @@ class TrainState:
         kw = {}
         if self.ops.name == "hip":
             kw["pad_rows"] = self._hip_rollout is not None
+            kw["out"] = self._pop_w      # stable pointer: hipGraph replay
         self._pop_w = self.ops.perturb(base, self.table, offsets, signs,
                                        cfg.sigma, sigma_vec=sigma_vec,
                                        out_dtype=out_dtype, **kw)

@@ -30,15 +30,15 @@ void des_launch_cartpole_reset(float*, int*, int64_t*, float*, uint8_t*,
                                float*, const int64_t*, int64_t, int,
                                hipStream_t);
 void des_launch_cartpole_step(float*, int*, int64_t*, const int*, float*,
-                              uint8_t*, float*, const int64_t*, int64_t, int,
-                              hipStream_t);
+                              uint8_t*, float*, const int64_t*,
+                              const int64_t*, int, hipStream_t);
 void des_launch_pong_reset(float*, float*, float*, float*, float*, float*,
                            int*, int*, int64_t*, uint8_t*, float*, uint8_t*,
                            float*, const int64_t*, int64_t, int, hipStream_t);
 void des_launch_pong_step(float*, float*, float*, float*, float*, float*,
                           int*, int*, int64_t*, uint8_t*, int, const int*,
-                          float*, uint8_t*, float*, const int64_t*, int64_t,
-                          int, hipStream_t);
+                          float*, uint8_t*, float*, const int64_t*,
+                          const int64_t*, int, hipStream_t);
 void des_launch_latent_reset(float*, int*, float*, uint8_t*, float*,
                              const int64_t*, int64_t, int, hipStream_t);
 void des_launch_latent_render(const float*, uint16_t*, int, int, int,
@@ -196,11 +196,12 @@ void cartpole_step(torch::Tensor state, torch::Tensor steps,
                    torch::Tensor reset_ctr, torch::Tensor actions,
                    torch::Tensor fitness, torch::Tensor alive,
                    torch::Tensor ep_len, torch::Tensor env_ids,
-                   int64_t seed_mix) {
+                   torch::Tensor seed_buf) {
   int P = state.size(0);
+  TORCH_CHECK(seed_buf.is_cuda() && seed_buf.dtype() == torch::kInt64);
   des_launch_cartpole_step(fptr(state), iptr(steps), lptr(reset_ctr),
                            iptr(actions), fptr(fitness), bptr(alive),
-                           fptr(ep_len), lptr(env_ids), seed_mix, P,
+                           fptr(ep_len), lptr(env_ids), lptr(seed_buf), P,
                            cur_stream());
 }
 
@@ -223,13 +224,14 @@ void pong_step(torch::Tensor bx, torch::Tensor by, torch::Tensor vx,
                torch::Tensor serve_ctr, torch::Tensor frames, int64_t slot,
                torch::Tensor actions, torch::Tensor fitness,
                torch::Tensor alive, torch::Tensor ep_len,
-               torch::Tensor env_ids, int64_t seed) {
+               torch::Tensor env_ids, torch::Tensor seed_buf) {
   int P = bx.size(0);
+  TORCH_CHECK(seed_buf.is_cuda() && seed_buf.dtype() == torch::kInt64);
   des_launch_pong_step(fptr(bx), fptr(by), fptr(vx), fptr(vy), fptr(py),
                        fptr(oy), iptr(score), iptr(points), lptr(serve_ctr),
                        bptr(frames), (int)slot, iptr(actions), fptr(fitness),
-                       bptr(alive), fptr(ep_len), lptr(env_ids), seed, P,
-                       cur_stream());
+                       bptr(alive), fptr(ep_len), lptr(env_ids),
+                       lptr(seed_buf), P, cur_stream());
 }
 
 void latent_reset(torch::Tensor z, torch::Tensor steps, torch::Tensor fitness,

@@ -53,9 +53,11 @@ __global__ void cartpole_step_kernel(float* __restrict__ state,
                                      uint8_t* __restrict__ alive,
                                      float* __restrict__ ep_len,
                                      const int64_t* __restrict__ env_ids,
-                                     int64_t seed_term, int P) {
+                                     const int64_t* __restrict__ seedp,
+                                     int P) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= P) return;
+  const int64_t seed_term = *seedp;  // device-read: hipGraph-replayable
   float x = state[i * 4], xd = state[i * 4 + 1], th = state[i * 4 + 2],
         td = state[i * 4 + 3];
   float force = actions[i] ? FORCE : -FORCE;
@@ -102,8 +104,8 @@ void des_launch_cartpole_reset(float* state, int* steps, int64_t* reset_ctr,
 void des_launch_cartpole_step(float* state, int* steps, int64_t* reset_ctr,
                               const int* actions, float* fitness,
                               uint8_t* alive, float* ep_len,
-                              const int64_t* env_ids, int64_t seed_term, int P,
-                              hipStream_t stream) {
+                              const int64_t* env_ids, const int64_t* seed_term,
+                              int P, hipStream_t stream) {
   int t = 256, g = (P + t - 1) / t;
   hipLaunchKernelGGL(cartpole_step_kernel, dim3(g), dim3(t), 0, stream, state,
                      steps, reset_ctr, actions, fitness, alive, ep_len,

@@ -78,10 +78,11 @@ __global__ void pong_step_kernel(PongState st, uint8_t* __restrict__ frames,
                                  uint8_t* __restrict__ alive,
                                  float* __restrict__ ep_len,
                                  const int64_t* __restrict__ env_ids,
-                                 int64_t seed, int P) {
+                                 const int64_t* __restrict__ seedp, int P) {
   int i = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   int lane = threadIdx.x & 63;
   if (i >= P) return;
+  const int64_t seed = *seedp;  // device-read: hipGraph-replayable
   int64_t id = env_ids[i];
   float bx = st.bx[i], by = st.by[i], vx = st.vx[i], vy = st.vy[i];
   float py = st.py[i], oy = st.oy[i];
@@ -164,8 +165,8 @@ void des_launch_pong_step(float* bx, float* by, float* vx, float* vy,
                           float* py, float* oy, int* score, int* points,
                           int64_t* serve_ctr, uint8_t* frames, int slot,
                           const int* actions, float* fitness, uint8_t* alive,
-                          float* ep_len, const int64_t* env_ids, int64_t seed,
-                          int P, hipStream_t stream) {
+                          float* ep_len, const int64_t* env_ids,
+                          const int64_t* seed, int P, hipStream_t stream) {
   PongState st{bx, by, vx, vy, py, oy, score, points, serve_ctr};
   int waves_per_block = 4;
   int g = (P + waves_per_block - 1) / waves_per_block;

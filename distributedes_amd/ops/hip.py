@@ -59,13 +59,15 @@ class HipOps:
     # K2
     def perturb(self, theta, table, offsets, signs, sigma,
                 sigma_vec=None, out_dtype=torch.float32, chunk=None,
-                pad_rows: bool = False):
+                pad_rows: bool = False, out=None):
         """pad_rows: allocate [P, pad16(D)] so every member row is 16B-aligned
-        for the vectorized policy-forward kernels (pad tail is unwritten)."""
+        for the vectorized policy-forward kernels (pad tail is unwritten).
+        out: reuse a persistent buffer (stable pointer for hipGraph replay)."""
         P = offsets.shape[0]
         D = theta.shape[0]
         stride = (D + 7) & ~7 if pad_rows else D
-        out = torch.empty(P, stride, dtype=out_dtype, device=theta.device)
+        if out is None or out.shape != (P, stride) or out.dtype != out_dtype:
+            out = torch.empty(P, stride, dtype=out_dtype, device=theta.device)
         self.ext.perturb(out, theta, table, offsets, signs, float(sigma),
                          sigma_vec if sigma_vec is not None else
                          torch.empty(0, device=theta.device))

@@ -100,6 +100,8 @@ class HipRollout:
         self.P = P
         z = lambda *s, dt=torch.float32: torch.zeros(*s, dtype=dt, device=dev)  # noqa: E731
         self._empty = torch.empty(0, device=dev)
+        self.seed_buf = torch.zeros(1, dtype=torch.int64, device=dev)
+        self._graphs = {}
         self.fitness = z(P)
         self.alive = z(P, dt=torch.uint8)
         self.ep_len = z(P)
@@ -167,6 +169,9 @@ class HipRollout:
                                     self.alive, self.ep_len, ids, gen_seed)
         self._ids = ids
         self._gen_seed = gen_seed
+        # the step kernels read their hash seed from device memory so a
+        # captured hipGraph replays correctly across generations
+        self.seed_buf.fill_(gen_seed * 0x10001 if e == "cartpole" else gen_seed)
 
     # ------------------------------------------------------------------ #
     def _forward(self, pop_w: torch.Tensor, normalizer, slot_newest: int):
@@ -233,14 +238,13 @@ class HipRollout:
         if e == "cartpole":
             self.ext.cartpole_step(self.st_state, self.st_steps, self.st_rctr,
                                    self.actions_i, self.fitness, self.alive,
-                                   self.ep_len, self._ids,
-                                   self._gen_seed * 0x10001)
+                                   self.ep_len, self._ids, self.seed_buf)
         elif e == "pong":
             self.ext.pong_step(self.pg_bx, self.pg_by, self.pg_vx, self.pg_vy,
                                self.pg_py, self.pg_oy, self.pg_score,
                                self.pg_points, self.pg_serve, self.frames,
                                slot, self.actions_i, self.fitness, self.alive,
-                               self.ep_len, self._ids, self._gen_seed)
+                               self.ep_len, self._ids, self.seed_buf)
         elif e in ("atari57_synth", "procgen_synth"):
             self.ext.latent_step(self.sy_z, self.sy_steps, self.actions_i,
                                  self.fitness, self.alive, self.ep_len,
@@ -252,21 +256,15 @@ class HipRollout:
                                    self.env.horizon)
 
     def _observe(self, normalizer):
-        """K12: fold current obs into the normalizer's fp64 delta buffers."""
+        """K12: fold current obs into the normalizer's fp64 delta buffers
+        (the count is added analytically in run(): graph-capture-safe)."""
         if not hasattr(normalizer, "_dsum"):
             return
         obs = self.st_state if self.env_name == "cartpole" else self.hu_s
         self.ext.normalizer_observe(obs, normalizer._dsum, normalizer._dsumsq)
-        normalizer._dcount += obs.shape[0]
 
     # ------------------------------------------------------------------ #
-    def run(self, pop_w: torch.Tensor, gen_seed: int, env_ids: torch.Tensor,
-            episode_steps: int, normalizer):
-        P = pop_w.shape[0]
-        if self.P != P:
-            self._alloc(P)
-        is_state_env = len(self.env.obs_shape) == 1
-        self._reset(gen_seed, env_ids)
+    def _loop(self, pop_w, episode_steps, normalizer, is_state_env):
         # pong ring: reset wrote the initial frame to every slot; step t
         # writes slot t % 4 and the forward sees slots in logical order
         for t in range(episode_steps):
@@ -275,4 +273,53 @@ class HipRollout:
                 self._observe(normalizer)
             self._forward(pop_w, normalizer, slot_newest=(t - 1) % 4)
             self._env_step(slot)
+
+    def run(self, pop_w: torch.Tensor, gen_seed: int, env_ids: torch.Tensor,
+            episode_steps: int, normalizer):
+        P = pop_w.shape[0]
+        if self.P != P:
+            self._alloc(P)
+        is_state_env = len(self.env.obs_shape) == 1
+        self._reset(gen_seed, env_ids)
+        use_graph = (getattr(self.cfg, "use_hip_graphs", False)
+                     and self.device.type == "cuda"
+                     and not getattr(self, "_graphs_broken", False))
+        if use_graph:
+            key = (episode_steps, pop_w.data_ptr(), id(normalizer))
+            g = self._graphs.get(key)
+            if g is None:
+                try:
+                    g = self._capture(pop_w, gen_seed, env_ids, episode_steps,
+                                      normalizer, is_state_env)
+                    self._graphs[key] = g
+                except Exception:
+                    self._graphs_broken = True
+                    use_graph = False
+            if use_graph:
+                g.replay()
+        if not use_graph:
+            self._loop(pop_w, episode_steps, normalizer, is_state_env)
+        if is_state_env and hasattr(normalizer, "_dcount"):
+            normalizer._dcount += P * episode_steps
         return self.fitness.clone(), P * episode_steps, float(self.ep_len.mean())
+
+    def _capture(self, pop_w, gen_seed, env_ids, episode_steps, normalizer,
+                 is_state_env):
+        """Capture one generation's kernel sequence into a hipGraph.
+        The warmup pass EXECUTES a full loop (torch requirement), so env +
+        normalizer-delta state is re-reset before capture; the replay then
+        performs the real generation."""
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            self._loop(pop_w, episode_steps, normalizer, is_state_env)
+        torch.cuda.current_stream().wait_stream(s)
+        # undo warmup side effects
+        self._reset(self._gen_seed, self._ids)
+        if hasattr(normalizer, "_dsum"):
+            normalizer._dsum.zero_()
+            normalizer._dsumsq.zero_()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._loop(pop_w, episode_steps, normalizer, is_state_env)
+        return g

@@ -57,8 +57,10 @@ class RunningNormalizer:
         mean = self.sum / self.count
         var = self.sumsq / self.count - mean * mean
         var = var.clamp_min(1e-8)
-        self._mean = mean.to(torch.float32)
-        self._inv_std = var.rsqrt().to(torch.float32)
+        # in-place: _mean/_inv_std pointers stay stable so captured hipGraphs
+        # (which bake the addresses into the MLP forward) see fresh stats
+        self._mean.copy_(mean.to(torch.float32))
+        self._inv_std.copy_(var.rsqrt().to(torch.float32))
 
     def normalize(self, obs: torch.Tensor) -> torch.Tensor:
         x = (obs.to(torch.float32) - self._mean) * self._inv_std
