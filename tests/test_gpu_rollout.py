@@ -162,12 +162,13 @@ def test_cartpole_kernel_matches_torch_env(ext):
     eplen = torch.zeros(P, device=dev)
     ext.cartpole_reset(state, steps, rctr, fit, alive, eplen, ids, seed * 0x10001)
     torch.testing.assert_close(state.cpu(), obs_ref, rtol=1e-6, atol=1e-7)
+    seed_buf = torch.tensor([seed * 0x10001], dtype=torch.int64, device=dev)
     torch.manual_seed(0)
     for t in range(120):
         a = torch.randint(0, 2, (P,))
         obs_ref, r, d = env.step(a)
         ext.cartpole_step(state, steps, rctr, a.to(torch.int32).cuda(), fit,
-                          alive, eplen, ids, seed * 0x10001)
+                          alive, eplen, ids, seed_buf)
         torch.testing.assert_close(state.cpu(), obs_ref, rtol=1e-4, atol=1e-4)
 
 
@@ -188,13 +189,14 @@ def test_pong_kernel_matches_torch_env(ext):
                    alive, eplen, ids, seed)
     torch.testing.assert_close(bx.cpu(), env.bx, rtol=1e-5, atol=1e-5)
     torch.testing.assert_close(by.cpu(), env.by, rtol=1e-5, atol=1e-5)
+    seed_buf = torch.tensor([seed], dtype=torch.int64, device=dev)
     total_r = torch.zeros(P)
     for t in range(200):
         a = torch.full((P,), t % 3, dtype=torch.int64)
         obs_ref, r, d = env.step(a)
         ext.pong_step(bx, by, vx, vy, py, oy, score, points, serve, frames,
                       t % 4, a.to(torch.int32).cuda(), fit, alive, eplen, ids,
-                      seed)
+                      seed_buf)
         total_r += r
         torch.testing.assert_close(by.cpu(), env.by, rtol=1e-3, atol=1e-3)
         torch.testing.assert_close(py.cpu(), env.player_y, rtol=1e-4, atol=1e-4)
