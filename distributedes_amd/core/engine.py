@@ -107,6 +107,7 @@ class TrainState:
         self.metrics = MetricsLogger(cfg.metrics_path, rank=self.rank,
                                      log_every=cfg.log_every)
         self._pop_w = None          # reusable [local_pop, D] perturbed weights
+        self._offs_cache = {}       # generation -> (offsets, signs) prefetch
 
         # GPU-native rollout path (K3/K5/K7/K8 kernels); torch loop otherwise
         self._hip_rollout = None
@@ -118,7 +119,12 @@ class TrainState:
 
     # ------------------------------------------------------------------ #
     def member_offsets_signs(self, generation: int):
-        """Offsets/signs for the LOCAL member slice, plus global arrays."""
+        """Offsets/signs for the LOCAL member slice.  Results are cached one
+        generation ahead: train_step derives generation g+1 on the HOST while
+        the C3/C4 collectives for generation g ride the side stream."""
+        cached = self._offs_cache.pop(generation, None)
+        if cached is not None:
+            return cached
         cfg = self.cfg
         offs, signs = noise_mod.member_offsets(
             cfg.seed, generation, cfg.pop, cfg.antithetic,
@@ -127,6 +133,10 @@ class TrainState:
         offs_t = torch.from_numpy(offs[lo:hi]).to(self.device)
         signs_t = torch.from_numpy(signs[lo:hi]).to(self.device)
         return offs_t, signs_t
+
+    def prefetch_offsets(self, generation: int):
+        if generation not in self._offs_cache:
+            self._offs_cache[generation] = self.member_offsets_signs(generation)
 
     def env_ids(self):
         """Global env-stream ids for local members (pair-shared if antithetic)."""
@@ -268,7 +278,17 @@ def train_step(state: TrainState) -> dict:
     pop_w = state.perturb_local(offsets, signs)                      # K2
     fitness, env_steps, ep_len = state.rollout(pop_w)                # K7/K3/K5/K8
 
-    global_fitness = state.comm.allgather_fitness(fitness)           # C3
+    # C3 rides the side HIP stream (overlap_comm) while the host derives the
+    # NEXT generation's member offsets (BASELINE: collectives overlapped with
+    # the next generation's noise draw)
+    if cfg.overlap_comm and state.device.type == "cuda" and state.comm.is_dist:
+        with state.comm.sidestream():
+            global_fitness = state.comm.allgather_fitness(fitness)   # C3
+        state.prefetch_offsets(state.generation + 1)
+        state.comm.join_sidestream()
+    else:
+        global_fitness = state.comm.allgather_fitness(fitness)       # C3
+        state.prefetch_offsets(state.generation + 1)
     if cfg.algo == "es":
         state.es_update(global_fitness, offsets, signs)              # K9/K10/C4
     else:
