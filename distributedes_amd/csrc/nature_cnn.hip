@@ -1,55 +1,98 @@
-// K5: conv_forward_pop (Nature-CNN) — per-member policy forward as three
-// kernels sized for their regimes:
+// K5: conv_forward_pop (Nature-CNN) — per-member policy forward, one fused
+// block-per-member kernel (BASELINE configs #3/#4).
 //
-// ONE block-per-member kernel (512 thr, ~77 KiB LDS -> 2 blocks/CU):
-//   convs: 8x8s4x32 -> 4x4s2x64 -> 3x3s1x64 on packed-pair v_dot2c_f32_bf16,
-//     weight tiles staged in 32-oc halves to conflict-free [k/4][oc][4] bf16
-//     LDS layouts; activations live in LDS only;
-//   fc 3136->512: the HBM-bound stage (94% of weights) streamed with 4-row
-//     ILP bf16x8 (16 B) loads per wave;
-//   head 512->A + argmax fused.
-// Because 2 blocks co-reside per CU, one block's VALU-bound conv phase
-// overlaps the other block's HBM-bound FC stream: step time ~ max(conv, fc)
-// rather than the sum.  FC floor at pop=8192/GPU:
-// 8192 * 3.2 MiB / 6.3 TBps ~= 4.3 ms.
+// ONE 512-thread (8-wave) block per member, ~64 KiB LDS -> 2 blocks/CU:
+//
+//   convs as MFMA implicit GEMM (v_mfma_f32_16x16x32_bf16):
+//     conv1  M=400(25 tiles) x N=32(2) x K=256   (im2col of the obs frames)
+//     conv2  M=96(6, 81 real) x N=64(4) x K=512  (im2col of a1, LDS)
+//     conv3  M=64(4, 49 real) x N=64(4) x K=576  (im2col of a2, LDS)
+//   Per K-step (32): all 512 threads cooperatively build the A-slab
+//   (im2col gather -> [M][32] bf16, 64-B rows) and B-slab ([N][32] from the
+//   member's flat [oc][k] weights: each lane's b128 fragment read is 8
+//   consecutive k of its oc row -> no transpose pass at all), then each
+//   wave issues one MFMA per owned 16x16 C-tile.  Fragment maps verified on
+//   hardware by tools/probe_mfma.hip:
+//     A: lane l -> A[l&15][(l>>4)*8+j]   B: lane l -> B[(l>>4)*8+j][l&15]
+//     D: lane l, reg r -> D[(l>>4)*4+r][l&15]
+//   Activations never leave LDS; relu+bias fused in the tile epilogues.
+//
+//   fc 3136->512 (94% of weights = the HBM floor): 4 rows of ILP per wave,
+//   16-B bf16x8 loads on v_dot2c_f32_bf16; head 512->A + argmax fused.
+//
+// With 2 blocks/CU one block's (now tiny) conv phase and the other block's
+// HBM-bound FC stream overlap; the step floor is the FC traffic:
+// pop-8192 x 3.2 MiB / 6.3 TB/s ~= 4.3 ms.
 #include "common.h"
 #include "gemv.h"
 
 namespace {
 
+typedef __attribute__((ext_vector_type(8))) short mfma_ab;
+typedef __attribute__((ext_vector_type(4))) float mfma_cd;
+
 // geometry (fixed, models/nature_cnn.py)
 constexpr int C_IN = 4, HW = 84;
-constexpr int C1 = 32, O1 = 20, K1 = 8, S1 = 4;   // conv1
-constexpr int C2 = 64, O2 = 9, K2 = 4, S2 = 2;    // conv2
-constexpr int C3 = 64, O3 = 7;                    // conv3 (3x3 s1)
-constexpr int FC_IN = C3 * O3 * O3;               // 3136
-constexpr int FC_OUT = 512;
-constexpr int MAX_A = 32;
+constexpr int O1 = 20, K1 = 8, S1 = 4;            // conv1 8x8 s4, out 32ch
+constexpr int O2 = 9, S2 = 2;                     // conv2 4x4 s2, out 64ch
+constexpr int O3 = 7;                             // conv3 3x3 s1, out 64ch
+constexpr int FC_IN = 3136, FC_OUT = 512, MAX_A = 32;
 
-// ---------------- kernel 1: convs ----------------
-// LDS carve (bytes, 16-aligned).  conv2/conv3 weights staged in 32-oc
-// halves so the whole carve stays < 80 KiB -> 2 blocks/CU (latency hiding).
-constexpr int W_ELEMS = 32 * 576;                      // 18432 (conv3 half)
-constexpr int OFF_W = 0;                               // bf16 [W_ELEMS]
-constexpr int OFF_A1 = OFF_W + W_ELEMS * 2;            // bf16 [32*20*20]
-constexpr int OFF_A2 = OFF_A1 + C1 * O1 * O1 * 2;      // bf16 [64*9*9]
-constexpr int OFF_T2 = OFF_A2 + ((C2 * O2 * O2 * 2 + 15) & ~15);  // int [512]
-constexpr int OFF_T3 = OFF_T2 + 512 * 4;               // int [576]
-constexpr int CONV_LDS = OFF_T3 + 576 * 4;
-// fc input (3136 bf16), fc output (512 f32) and head logits overlay the a1
-// carve (dead after conv2): 6272 + 2048 + 128 B < 25600 B
-constexpr int OFF_A3 = OFF_A1;
-constexpr int OFF_FC = OFF_A3 + FC_IN * 2;
-constexpr int OFF_HD = OFF_FC + FC_OUT * 4;
+// LDS carve (bytes, 16-aligned): A-slab (max conv1 400x32 bf16), B-slab
+// (64x32), a1, a2; fc input/out + head overlay a1 (dead after conv2)
+constexpr int OFF_SA = 0;                         // [400][32] bf16 = 25600
+constexpr int OFF_SB = OFF_SA + 400 * 64;         // [64][32] bf16 = 4096
+constexpr int OFF_A1 = OFF_SB + 64 * 64;          // [32][400] bf16 = 25600
+constexpr int OFF_A2 = OFF_A1 + 32 * 400 * 2;     // [64][81] bf16 = 10368
+constexpr int CONV_LDS = (OFF_A2 + 64 * 81 * 2 + 15) & ~15;
+constexpr int OFF_A3 = OFF_A1;                    // bf16 fc input [3136]
+constexpr int OFF_FC = OFF_A3 + FC_IN * 2;        // f32 [512]
+constexpr int OFF_HD = OFF_FC + FC_OUT * 4;       // f32 [MAX_A]
 static_assert(OFF_HD + MAX_A * 4 <= OFF_A2, "fc overlay exceeds a1 carve");
 
-template <typename IT>
-__device__ __forceinline__ float obs_val(IT v);
-template <>
-__device__ __forceinline__ float obs_val<uint8_t>(uint8_t v) { return (float)v; }
-template <>
-__device__ __forceinline__ float obs_val<uint16_t>(uint16_t v) {
-  return des::bf16_to_f32(v);
+// one MFMA step over every tile a wave owns (MAXT static -> regs, rule 20)
+template <int MAXT>
+__device__ __forceinline__ void mfma_tiles(const uint16_t* sa,
+                                           const uint16_t* sb,
+                                           mfma_cd (&acc)[MAXT], int wid,
+                                           int lane, int ntiles, int ncols) {
+#pragma unroll
+  for (int u = 0; u < MAXT; ++u) {
+    int t = wid + 8 * u;
+    if (t < ntiles) {
+      int mt = t / ncols, nt = t % ncols;
+      mfma_ab af = *reinterpret_cast<const mfma_ab*>(
+          sa + (mt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
+      mfma_ab bf = *reinterpret_cast<const mfma_ab*>(
+          sb + (nt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
+      acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[u], 0, 0, 0);
+    }
+  }
+}
+
+// tile epilogue: bias + relu + scatter into [oc][M_real] LDS layout
+template <int MAXT>
+__device__ __forceinline__ void tile_epilogue(const mfma_cd (&acc)[MAXT],
+                                              const uint16_t* bias,
+                                              uint16_t* out, int m_real,
+                                              int wid, int lane, int ntiles,
+                                              int ncols) {
+#pragma unroll
+  for (int u = 0; u < MAXT; ++u) {
+    int t = wid + 8 * u;
+    if (t < ntiles) {
+      int mt = t / ncols, nt = t % ncols;
+      int n = nt * 16 + (lane & 15);
+      int m0 = mt * 16 + ((lane >> 4) << 2);
+      float b = des::bf16_to_f32(bias[n]);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int m = m0 + r;
+        if (m < m_real)
+          out[n * m_real + m] = des::f32_to_bf16(fmaxf(acc[u][r] + b, 0.f));
+      }
+    }
+  }
 }
 
 template <typename IT>
@@ -57,15 +100,14 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
     const uint16_t* __restrict__ popw, int64_t row_stride,
     const IT* __restrict__ frames,     // [P][4][84][84]
     const int* __restrict__ slot_map,  // [4] logical->physical slot
-    int* __restrict__ actions, float* __restrict__ logits_out, int A,
-    int P, const int64_t* __restrict__ offs) {
+    int* __restrict__ actions, float* __restrict__ logits_out, int A, int P,
+    const int64_t* __restrict__ offs) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  uint16_t* w_lds = reinterpret_cast<uint16_t*>(smem + OFF_W);
+  uint16_t* sa = reinterpret_cast<uint16_t*>(smem + OFF_SA);
+  uint16_t* sb = reinterpret_cast<uint16_t*>(smem + OFF_SB);
   uint16_t* a1 = reinterpret_cast<uint16_t*>(smem + OFF_A1);
   uint16_t* a2 = reinterpret_cast<uint16_t*>(smem + OFF_A2);
-  int* t2 = reinterpret_cast<int*>(smem + OFF_T2);
-  int* t3 = reinterpret_cast<int*>(smem + OFF_T3);
-  uint16_t* a3 = reinterpret_cast<uint16_t*>(smem + OFF_A3);   // fc input
+  uint16_t* a3 = reinterpret_cast<uint16_t*>(smem + OFF_A3);
   float* fcb = reinterpret_cast<float*>(smem + OFF_FC);
   float* hd = reinterpret_cast<float*>(smem + OFF_HD);
 
@@ -77,184 +119,113 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
   const int s0 = slot_map[0], s1m = slot_map[1], s2m = slot_map[2],
             s3 = slot_map[3];
 
-  // separable activation-offset tables: a1 index = base(pos) + t2[icc]
-  for (int e = tid; e < 512; e += 512) {
-    int ic = e >> 4, ky = (e >> 2) & 3, kx = e & 3;
-    t2[e] = ic * (O1 * O1) + ky * O1 + kx;
-  }
-  for (int e = tid; e < 576; e += 512) {
-    int ic = e / 9, r = e % 9;
-    t3[e] = ic * (O2 * O2) + (r / 3) * O2 + (r % 3);
-  }
-  // stage conv1 weights as [ic][ky][oc][kx8]: per-oc b128 gives 4 packed
-  // kx-pairs feeding v_dot2c_f32_bf16 directly
+  // ======================= conv1: 400 x 32 x 256 =========================
   {
     const uint16_t* w1 = w + offs[0];
-    for (int e = tid; e < C1 * 256; e += 512) {
-      int oc = e >> 8, rem = e & 255, icky = rem >> 3, kx = e & 7;
-      w_lds[(icky * C1 + oc) * K1 + kx] = w1[e];
+    mfma_cd acc[7];
+#pragma unroll
+    for (int u = 0; u < 7; ++u) acc[u] = 0.f;  // ext-vector splat
+    for (int ks = 0; ks < 8; ++ks) {   // k-step 32 = one ic, 4 ky rows
+      const int ic = ks >> 1, ky0 = (ks & 1) << 2;
+      const int slot = (ic == 0 ? s0 : ic == 1 ? s1m : ic == 2 ? s2m : s3);
+      const IT* plane = obs + slot * (HW * HW);
+      // A-slab: unit = (position m, ky row) -> 8 px -> 8 bf16 (16 B)
+      for (int e = tid; e < 400 * 4; e += 512) {
+        int m = e >> 2, kyl = e & 3;
+        int oy = m / O1, ox = m % O1;
+        const IT* row = plane + (oy * S1 + ky0 + kyl) * HW + ox * S1;
+        uint4 pk;
+        if constexpr (sizeof(IT) == 1) {
+          uint32_t d0 = *reinterpret_cast<const uint32_t*>(row);
+          uint32_t d1 = *reinterpret_cast<const uint32_t*>(row + 4);
+          pk.x = __builtin_bit_cast(uint32_t,
+                 des::pack_bf16((float)(d0 & 0xFF), (float)((d0 >> 8) & 0xFF)));
+          pk.y = __builtin_bit_cast(uint32_t,
+                 des::pack_bf16((float)((d0 >> 16) & 0xFF), (float)(d0 >> 24)));
+          pk.z = __builtin_bit_cast(uint32_t,
+                 des::pack_bf16((float)(d1 & 0xFF), (float)((d1 >> 8) & 0xFF)));
+          pk.w = __builtin_bit_cast(uint32_t,
+                 des::pack_bf16((float)((d1 >> 16) & 0xFF), (float)(d1 >> 24)));
+        } else {
+          uint2 d0 = *reinterpret_cast<const uint2*>(row);
+          uint2 d1 = *reinterpret_cast<const uint2*>(row + 4);
+          pk = {d0.x, d0.y, d1.x, d1.y};
+        }
+        *reinterpret_cast<uint4*>(sa + m * 32 + kyl * 8) = pk;
+      }
+      // B-slab: [32 oc][32 k] from w1 rows (K = 256)
+      for (int e = tid; e < 32 * 32; e += 512) {
+        int n = e >> 5, kk = e & 31;
+        sb[n * 32 + kk] = w1[n * 256 + ks * 32 + kk];
+      }
+      __syncthreads();
+      mfma_tiles<7>(sa, sb, acc, wid, lane, 50, 2);
+      __syncthreads();
     }
+    tile_epilogue<7>(acc, w + offs[1], a1, 400, wid, lane, 50, 2);
   }
   __syncthreads();
 
-  // ---- conv1: thread = position (400 of 512); packed-pair dot2 ---------
+  // ======================= conv2: 96 x 64 x 512 ==========================
   {
-    const int pos = tid;
-    const int oy = pos / O1, ox = pos % O1;
-    const uint16_t* b1 = w + offs[1];
-    float acc[C1];
-    if (pos < O1 * O1) {
+    const uint16_t* w2 = w + offs[2];
+    mfma_cd acc[3];
 #pragma unroll
-      for (int k = 0; k < C1; ++k) acc[k] = 0.f;
-      for (int ic = 0; ic < C_IN; ++ic) {
-        const int slot = (ic == 0 ? s0 : ic == 1 ? s1m : ic == 2 ? s2m : s3);
-        const IT* plane = obs + slot * (HW * HW);
-        for (int ky = 0; ky < K1; ++ky) {
-          const IT* row = plane + (oy * S1 + ky) * HW + ox * S1;
-          // 8 pixels -> 4 packed bf16 pairs
-          des::bf16x2 pxp[4];
-          if constexpr (sizeof(IT) == 1) {           // u8 frames (dword loads)
-            uint32_t d0 = *reinterpret_cast<const uint32_t*>(row);
-            uint32_t d1 = *reinterpret_cast<const uint32_t*>(row + 4);
-#pragma unroll
-            for (int j = 0; j < 2; ++j) {
-              uint32_t d = j ? d1 : d0;
-              pxp[2 * j] = des::pack_bf16((float)(d & 0xFF),
-                                          (float)((d >> 8) & 0xFF));
-              pxp[2 * j + 1] = des::pack_bf16((float)((d >> 16) & 0xFF),
-                                              (float)(d >> 24));
-            }
-          } else {                                   // bf16 obs: already pairs
-            uint2 d0 = *reinterpret_cast<const uint2*>(row);
-            uint2 d1 = *reinterpret_cast<const uint2*>(row + 4);
-            pxp[0] = __builtin_bit_cast(des::bf16x2, d0.x);
-            pxp[1] = __builtin_bit_cast(des::bf16x2, d0.y);
-            pxp[2] = __builtin_bit_cast(des::bf16x2, d1.x);
-            pxp[3] = __builtin_bit_cast(des::bf16x2, d1.y);
-          }
-          const uint16_t* wrow = w_lds + ((ic * K1 + ky) * C1) * K1;
-#pragma unroll
-          for (int oc = 0; oc < C1; ++oc) {
-            uint4 wb = *reinterpret_cast<const uint4*>(wrow + oc * K1);
-            acc[oc] = des::dot2(pxp[0], __builtin_bit_cast(des::bf16x2, wb.x),
-                                acc[oc]);
-            acc[oc] = des::dot2(pxp[1], __builtin_bit_cast(des::bf16x2, wb.y),
-                                acc[oc]);
-            acc[oc] = des::dot2(pxp[2], __builtin_bit_cast(des::bf16x2, wb.z),
-                                acc[oc]);
-            acc[oc] = des::dot2(pxp[3], __builtin_bit_cast(des::bf16x2, wb.w),
-                                acc[oc]);
-          }
+    for (int u = 0; u < 3; ++u) acc[u] = 0.f;  // ext-vector splat
+    for (int ks = 0; ks < 16; ++ks) {  // k-step 32 = 2 ic x 4 ky x 4 kx
+      // A-slab: unit = (m, icloc, ky) -> 4 contiguous a1 x-values (8 B)
+      for (int e = tid; e < 96 * 8; e += 512) {
+        int m = e >> 3, icl = (e >> 2) & 1, ky = e & 3;
+        uint2 v = {0u, 0u};
+        if (m < O2 * O2) {
+          int oy = m / O2, ox = m % O2;
+          const uint16_t* ap = a1 + (ks * 2 + icl) * 400 +
+                               (oy * S2 + ky) * O1 + ox * S2;
+          v.x = *reinterpret_cast<const uint32_t*>(ap);       // even index
+          v.y = *reinterpret_cast<const uint32_t*>(ap + 2);
         }
+        *reinterpret_cast<uint2*>(sa + m * 32 + icl * 16 + ky * 4) = v;
       }
-#pragma unroll
-      for (int oc = 0; oc < C1; ++oc) {
-        float v = fmaxf(acc[oc] + des::load1(b1 + oc), 0.f);
-        a1[oc * (O1 * O1) + pos] = des::f32_to_bf16(v);
+      // B-slab: [64 oc][32 k] (K = 512)
+      for (int e = tid; e < 64 * 32; e += 512) {
+        int n = e >> 5, kk = e & 31;
+        sb[n * 32 + kk] = w2[n * 512 + ks * 32 + kk];
       }
+      __syncthreads();
+      mfma_tiles<3>(sa, sb, acc, wid, lane, 24, 4);
+      __syncthreads();
     }
+    tile_epilogue<3>(acc, w + offs[3], a2, 81, wid, lane, 24, 4);
   }
   __syncthreads();
 
-  // ---- conv2 in two 32-oc halves: lane&31 = oc, lane>>5 = pos stream ----
-  // q = (ic, ky); the [q][oc][kx4] tile makes w a uint2 of 2 packed
-  // kx-pairs, and a1's kx run is contiguous (even index -> aligned b32
-  // pair reads) -> 2 LDS dwords + 2 dot2 per 4 MACs per position
-  for (int half = 0; half < 2; ++half) {
-    const uint16_t* w2 = w + offs[2];
-    for (int e = tid; e < 32 * 512; e += 512) {
-      int oc = e >> 9, icc = e & 511;
-      w_lds[(icc >> 2) * 128 + oc * 4 + (icc & 3)] =
-          w2[(half * 32 + oc) * 512 + icc];
-    }
-    __syncthreads();
-    {
-      const int oc = lane & 31;
-      const uint16_t* b2 = w + offs[3];
-      const float bias = des::load1(b2 + half * 32 + oc);
-      const int pb = (wid * 2 + (lane >> 5)) * 8;  // 16 streams x 8 = 128 >= 81
-      float acc[8];
-      int pbase[8];
-#pragma unroll
-      for (int p = 0; p < 8; ++p) {
-        acc[p] = 0.f;
-        int pos = pb + p;
-        if (pos >= O2 * O2) pos = O2 * O2 - 1;  // clamp; result discarded
-        pbase[p] = (pos / O2) * S2 * O1 + (pos % O2) * S2;
-      }
-      for (int q = 0; q < 128; ++q) {
-        uint2 wp = *reinterpret_cast<const uint2*>(w_lds + q * 128 + oc * 4);
-        const int ab = t2[q << 2];        // ic*400 + ky*20 (kx = 0), even
-#pragma unroll
-        for (int p = 0; p < 8; ++p) {
-          const uint16_t* ap = a1 + pbase[p] + ab;
-          uint32_t a01 = *reinterpret_cast<const uint32_t*>(ap);
-          uint32_t a23 = *reinterpret_cast<const uint32_t*>(ap + 2);
-          acc[p] = des::dot2(__builtin_bit_cast(des::bf16x2, a01),
-                             __builtin_bit_cast(des::bf16x2, wp.x), acc[p]);
-          acc[p] = des::dot2(__builtin_bit_cast(des::bf16x2, a23),
-                             __builtin_bit_cast(des::bf16x2, wp.y), acc[p]);
-        }
-      }
-#pragma unroll
-      for (int p = 0; p < 8; ++p) {
-        int pos = pb + p;
-        if (pos < O2 * O2)
-          a2[(half * 32 + oc) * (O2 * O2) + pos] =
-              des::f32_to_bf16(fmaxf(acc[p] + bias, 0.f));
-      }
-    }
-    __syncthreads();
-  }
-
-  // ---- conv3: two oc-halves; output = fc input tile in LDS --------------
+  // ======================= conv3: 64 x 64 x 576 ==========================
   {
     const uint16_t* w3 = w + offs[4];
-    const uint16_t* b3 = w + offs[5];
-    for (int half = 0; half < 2; ++half) {
-      // stage [icc/4][oc32][4] for ocs [half*32, half*32+32)
-      for (int e = tid; e < 32 * 576; e += 512) {
-        int oc = e / 576, icc = e % 576;
-        w_lds[(icc >> 2) * 128 + oc * 4 + (icc & 3)] =
-            w3[(half * 32 + oc) * 576 + icc];
+    mfma_cd acc[2];
+#pragma unroll
+    for (int u = 0; u < 2; ++u) acc[u] = 0.f;  // ext-vector splat
+    for (int ks = 0; ks < 18; ++ks) {  // 576 = 18 x 32 exactly
+      for (int e = tid; e < 64 * 32; e += 512) {
+        int m = e >> 5, kk = e & 31;
+        int icc = ks * 32 + kk;
+        int ic = icc / 9, rm = icc - ic * 9, ky = rm / 3, kx = rm - ky * 3;
+        uint16_t v = 0;
+        if (m < O3 * O3) {
+          int oy = m / O3, ox = m % O3;
+          v = a2[ic * 81 + (oy + ky) * O2 + ox + kx];
+        }
+        sa[m * 32 + kk] = v;
+        sb[m * 32 + kk] = w3[m * 576 + icc];  // same e-split covers 64 oc
       }
       __syncthreads();
-      // lane&31 = oc-in-half; lane>>5 selects position stream (2 streams)
-      const int oc = (lane & 31);
-      const float bias = des::load1(b3 + half * 32 + oc);
-      for (int pb = (wid * 2 + (lane >> 5)) * 4; pb < O3 * O3; pb += 64) {
-        float acc[4];
-        int pbase[4];
-#pragma unroll
-        for (int p = 0; p < 4; ++p) {
-          acc[p] = 0.f;
-          int pos = pb + p;
-          if (pos >= O3 * O3) pos = O3 * O3 - 1;
-          pbase[p] = (pos / O3) * O2 + (pos % O3);
-        }
-        for (int icc = 0; icc < 576; icc += 4) {
-          float4 w4 = des::load4(w_lds + (icc >> 2) * 128 + oc * 4);
-          const int4 off = *reinterpret_cast<const int4*>(t3 + icc);
-#pragma unroll
-          for (int p = 0; p < 4; ++p) {
-            float a0 = des::bf16_to_f32(a2[pbase[p] + off.x]);
-            float b0 = des::bf16_to_f32(a2[pbase[p] + off.y]);
-            float c0 = des::bf16_to_f32(a2[pbase[p] + off.z]);
-            float d0 = des::bf16_to_f32(a2[pbase[p] + off.w]);
-            acc[p] += a0 * w4.x + b0 * w4.y + c0 * w4.z + d0 * w4.w;
-          }
-        }
-#pragma unroll
-        for (int p = 0; p < 4; ++p) {
-          int pos = pb + p;
-          if (pos < O3 * O3)
-            a3[(half * 32 + oc) * (O3 * O3) + pos] =
-                des::f32_to_bf16(fmaxf(acc[p] + bias, 0.f));
-        }
-      }
+      mfma_tiles<2>(sa, sb, acc, wid, lane, 16, 4);
       __syncthreads();
     }
+    // a2 fully consumed above; a3 overlays the (dead) a1 region
+    tile_epilogue<2>(acc, w + offs[5], a3, 49, wid, lane, 16, 4);
   }
+  __syncthreads();
 
   // ---- fc 3136 -> 512 (relu): HBM-bound, 4 rows of ILP per wave ---------
   {
