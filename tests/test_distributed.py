@@ -38,7 +38,17 @@ def _worker(rank, world, port, fn_name, out_q):
             dist.destroy_process_group()
 
 
-def _spawn(fn_name, world=2, port=29600):
+def _free_port():
+    import socket
+    sk = socket.socket()
+    sk.bind(("127.0.0.1", 0))
+    port = sk.getsockname()[1]
+    sk.close()
+    return port
+
+
+def _spawn(fn_name, world=2, port=None):
+    port = port or _free_port()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     procs = [ctx.Process(target=_worker, args=(r, world, port, fn_name, q))
@@ -85,7 +95,7 @@ def _body_resume(rank, world):
 
 
 def test_ws2_ranks_bitwise_identical_and_match_ws1():
-    res = _spawn("_body_train3", world=2, port=29611)
+    res = _spawn("_body_train3", world=2)
     # ranks agree bitwise (C4 all-reduce produces the same bits everywhere)
     assert torch.equal(res[0]["theta"], res[1]["theta"])
     assert res[0]["rets"] == res[1]["rets"]
@@ -107,5 +117,5 @@ def test_ws2_ranks_bitwise_identical_and_match_ws1():
 
 
 def test_ws2_resume_bit_equivalence():
-    res = _spawn("_body_resume", world=2, port=29627)
+    res = _spawn("_body_resume", world=2)
     assert res[0]["same"] and res[1]["same"]

@@ -73,12 +73,24 @@ def _resume_phase(rank, world, port, ckpt, out_q):
             dist.destroy_process_group()
 
 
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
 @pytest.mark.timeout(180)
 def test_rank_death_then_checkpoint_resume(tmp_path):
     ckpt = str(tmp_path / "g2.pt")
+    _free_port._p1 = _free_port()
+    _free_port._p2 = _free_port()
+    _free_port._p3 = _free_port()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_crash_phase, args=(r, 2, 29771, ckpt, q))
+    procs = [ctx.Process(target=_crash_phase, args=(r, 2, _free_port._p1, ckpt, q))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -92,7 +104,7 @@ def test_rank_death_then_checkpoint_resume(tmp_path):
 
     # fresh world resumes from the generation-2 checkpoint
     q2 = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_resume_phase, args=(r, 2, 29783, ckpt, q2))
+    procs = [ctx.Process(target=_resume_phase, args=(r, 2, _free_port._p2, ckpt, q2))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -115,7 +127,7 @@ def test_rank_death_then_checkpoint_resume(tmp_path):
     # reassociation), so instead assert the resumed world matches a second
     # resumed world (determinism of recovery itself)
     q3 = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_resume_phase, args=(r, 2, 29791, ckpt, q3))
+    procs = [ctx.Process(target=_resume_phase, args=(r, 2, _free_port._p3, ckpt, q3))
              for r in range(2)]
     for p in procs:
         p.start()
