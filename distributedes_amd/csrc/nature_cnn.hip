@@ -30,6 +30,7 @@ namespace {
 
 typedef __attribute__((ext_vector_type(8))) short mfma_ab;
 typedef __attribute__((ext_vector_type(4))) float mfma_cd;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;  // nt-loadable
 
 // geometry (fixed, models/nature_cnn.py)
 constexpr int C_IN = 4, HW = 84;
@@ -237,10 +238,16 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
       float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
       for (int ib = lane * 8; ib < FC_IN; ib += 64 * 8) {
         uint4 xp = *reinterpret_cast<const uint4*>(a3 + ib);
-        uint4 wa = *reinterpret_cast<const uint4*>(w0 + ib);
-        uint4 wb = *reinterpret_cast<const uint4*>(w0 + FC_IN + ib);
-        uint4 wc = *reinterpret_cast<const uint4*>(w0 + 2 * FC_IN + ib);
-        uint4 wd = *reinterpret_cast<const uint4*>(w0 + 3 * FC_IN + ib);
+        // nt: weights are read exactly once per step by one CU -> bypass
+        // caches (microarch nt-weights row: decode-like stream)
+        u32x4 wa = __builtin_nontemporal_load(
+            reinterpret_cast<const u32x4*>(w0 + ib));
+        u32x4 wb = __builtin_nontemporal_load(
+            reinterpret_cast<const u32x4*>(w0 + FC_IN + ib));
+        u32x4 wc = __builtin_nontemporal_load(
+            reinterpret_cast<const u32x4*>(w0 + 2 * FC_IN + ib));
+        u32x4 wd = __builtin_nontemporal_load(
+            reinterpret_cast<const u32x4*>(w0 + 3 * FC_IN + ib));
         const uint32_t* xq = reinterpret_cast<const uint32_t*>(&xp);
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
