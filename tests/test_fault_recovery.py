@@ -29,7 +29,7 @@ def _init(rank, world, port):
                       WORLD_SIZE=str(world), RANK=str(rank),
                       LOCAL_RANK=str(rank))
     dist.init_process_group("gloo", rank=rank, world_size=world,
-                            timeout=datetime.timedelta(seconds=20))
+                            timeout=datetime.timedelta(seconds=45))
 
 
 def _crash_phase(rank, world, port, ckpt, out_q):
