@@ -55,4 +55,52 @@ __device__ void gemv_rows(const WT* __restrict__ w, const WT* __restrict__ b,
   }
 }
 
+// 4-rows-of-ILP variant: 4 independent weight streams per wave hide load
+// latency (row index clamped so partial last blocks stay in-bounds and the
+// overflow lanes simply discard their result)
+template <typename WT, int ACT>
+__device__ void gemv_rows4(const WT* __restrict__ w, const WT* __restrict__ b,
+                           const float* __restrict__ x, float* __restrict__ y,
+                           int in_dim, int out_dim, int64_t stride, int lane,
+                           int wid, int nwaves) {
+  int full = in_dim & ~3;
+  for (int r0 = wid * 4; r0 < out_dim; r0 += nwaves * 4) {
+    int rc = out_dim - 1;
+    const WT* w0 = w + (int64_t)r0 * stride;
+    const WT* w1 = w + (int64_t)min(r0 + 1, rc) * stride;
+    const WT* w2 = w + (int64_t)min(r0 + 2, rc) * stride;
+    const WT* w3 = w + (int64_t)min(r0 + 3, rc) * stride;
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    for (int i = lane * 4; i < full; i += 64 * 4) {
+      float x0 = x[i], x1 = x[i + 1], x2 = x[i + 2], x3 = x[i + 3];
+      float4 v0 = load4(w0 + i);
+      float4 v1 = load4(w1 + i);
+      float4 v2 = load4(w2 + i);
+      float4 v3 = load4(w3 + i);
+      a0 += v0.x * x0 + v0.y * x1 + v0.z * x2 + v0.w * x3;
+      a1 += v1.x * x0 + v1.y * x1 + v1.z * x2 + v1.w * x3;
+      a2 += v2.x * x0 + v2.y * x1 + v2.z * x2 + v2.w * x3;
+      a3 += v3.x * x0 + v3.y * x1 + v3.z * x2 + v3.w * x3;
+    }
+    for (int i = full + lane; i < in_dim; i += 64) {
+      float xv = x[i];
+      a0 += load1(w0 + i) * xv;
+      a1 += load1(w1 + i) * xv;
+      a2 += load1(w2 + i) * xv;
+      a3 += load1(w3 + i) * xv;
+    }
+    a0 = wave_reduce_sum_f(a0);
+    a1 = wave_reduce_sum_f(a1);
+    a2 = wave_reduce_sum_f(a2);
+    a3 = wave_reduce_sum_f(a3);
+    if (lane < 4 && r0 + lane < out_dim) {
+      float acc = lane == 0 ? a0 : lane == 1 ? a1 : lane == 2 ? a2 : a3;
+      acc += load1(b + r0 + lane);
+      if (ACT == 1) acc = tanhf(acc);
+      if (ACT == 2) acc = fmaxf(acc, 0.f);
+      y[r0 + lane] = acc;
+    }
+  }
+}
+
 }  // namespace des

@@ -44,19 +44,20 @@ __global__ __launch_bounds__(256) void mlp_pop_forward_kernel(
   }
   __syncthreads();
 
-  des::gemv_rows<WT, 1>(w + ow1, w + ob1, xbuf, ybuf, in_dim, h1, in_dim,
-                        lane, wid, 4);
+  des::gemv_rows4<WT, 1>(w + ow1, w + ob1, xbuf, ybuf, in_dim, h1, in_dim,
+                         lane, wid, 4);
   __syncthreads();
   const float* x2 = ybuf;
   float* y2 = xbuf;
   int last_dim = h1;
   if (h2 > 0) {
-    des::gemv_rows<WT, 1>(w + ow2, w + ob2, ybuf, xbuf, h1, h2, h1, lane, wid, 4);
+    des::gemv_rows4<WT, 1>(w + ow2, w + ob2, ybuf, xbuf, h1, h2, h1, lane,
+                           wid, 4);
     __syncthreads();
     x2 = xbuf; y2 = ybuf; last_dim = h2;
   }
-  des::gemv_rows<WT, 0>(w + owh, w + obh, x2, y2, last_dim, A, last_dim, lane,
-                        wid, 4);
+  des::gemv_rows4<WT, 0>(w + owh, w + obh, x2, y2, last_dim, A, last_dim,
+                         lane, wid, 4);
   __syncthreads();
 
   if (logits_out)

@@ -107,13 +107,19 @@ class Comm:
                               dtype=local_fit.dtype, device=local_fit.device)
             dist.all_gather_into_tensor(out, local_fit)
             return out
-        parts = [torch.empty_like(local_fit) for _ in range(self.world)]
-        dist.all_gather(parts, local_fit)
-        return torch.cat(parts)
+        src = local_fit.cpu() if local_fit.device.type != "cpu" else local_fit
+        parts = [torch.empty_like(src) for _ in range(self.world)]
+        dist.all_gather(parts, src)
+        return torch.cat(parts).to(local_fit.device)
 
     def allreduce_(self, t: torch.Tensor, avg: bool = False):     # C4/C5/C6
         if self.is_dist:
-            dist.all_reduce(t)
+            if self.backend != "nccl" and t.device.type != "cpu":
+                cpu = t.cpu()
+                dist.all_reduce(cpu)
+                t.copy_(cpu)
+            else:
+                dist.all_reduce(t)
             if avg:
                 t.div_(self.world)
         return t
