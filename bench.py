@@ -36,6 +36,9 @@ def main():
     ap.add_argument("--ops", default=None, choices=[None, "torch_ref", "hip"],
                     help="ops backend override (self-baseline: torch_ref)")
     ap.add_argument("--table", type=int, default=None, help="noise table floats")
+    ap.add_argument("--profile", default=None, metavar="DIR",
+                    help="write a torch.profiler chrome trace of the timed "
+                         "steps into DIR (SURVEY.md §5.1 tracing hook)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -70,6 +73,11 @@ def main():
     state.comm.barrier()
     if on_gpu:
         torch.cuda.synchronize()
+    prof = None
+    if args.profile:
+        from torch.profiler import ProfilerActivity, profile
+        prof = profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA])
+        prof.__enter__()
     t0 = time.perf_counter()
     env_steps = 0
     for _ in range(args.steps):
@@ -77,6 +85,11 @@ def main():
         env_steps += s["env_steps"]
     if on_gpu:
         torch.cuda.synchronize()
+    if prof is not None:
+        prof.__exit__(None, None, None)
+        os.makedirs(args.profile, exist_ok=True)
+        prof.export_chrome_trace(os.path.join(
+            args.profile, f"trace_rank{rank}.json"))
     state.comm.barrier()
     elapsed = time.perf_counter() - t0
 
