@@ -36,6 +36,10 @@ def main():
     ap.add_argument("--ops", default=None, choices=[None, "torch_ref", "hip"],
                     help="ops backend override (self-baseline: torch_ref)")
     ap.add_argument("--table", type=int, default=None, help="noise table floats")
+    ap.add_argument("--backend", default=None, choices=[None, "nccl", "gloo"],
+                    help="process-group backend override (default: nccl on "
+                         "GPU; gloo lets N ranks share one GPU for plumbing "
+                         "dry-runs)")
     ap.add_argument("--profile", default=None, metavar="DIR",
                     help="write a torch.profiler chrome trace of the timed "
                          "steps into DIR (SURVEY.md §5.1 tracing hook)")
@@ -47,7 +51,7 @@ def main():
 
     on_gpu = torch.cuda.is_available()
     if on_gpu:
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(min(local_rank, torch.cuda.device_count() - 1))
 
     overrides = dict(generations=10 ** 9, checkpoint_every=0,
                      metrics_path=None, log_every=10 ** 9)
@@ -57,6 +61,8 @@ def main():
         overrides["episode_steps"] = args.episode_steps
     if args.table:
         overrides["noise_table_size"] = args.table
+    if args.backend:
+        overrides["backend"] = args.backend
     if not on_gpu:
         # CPU smoke only (the driver benches on a GPU box): shrink hard
         overrides.setdefault("pop", 8)
@@ -96,6 +102,7 @@ def main():
     # MAX over ranks (slowest rank defines job time)
     el = torch.tensor([elapsed], device=state.device
                       if state.comm.backend == "nccl" else "cpu")
+    del s
     if state.comm.is_dist:
         import torch.distributed as dist
         dist.all_reduce(el, op=dist.ReduceOp.MAX)
