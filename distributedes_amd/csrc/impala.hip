@@ -15,6 +15,15 @@
 
 namespace {
 
+typedef __attribute__((ext_vector_type(8))) short mfma_ab;
+typedef __attribute__((ext_vector_type(4))) float mfma_cd;
+
+// 3x3 pad-1 conv as MFMA implicit GEMM per spatial tile:
+//   M = TS*TS positions, N = CO, K = CI*9 (zero-padded to 32-multiples).
+// Per K-step all threads build the A-slab (im2col from the halo-staged
+// input tile) and B-slab (direct [co][k] rows of the member's weights),
+// then each wave runs one v_mfma_f32_16x16x32_bf16 per owned 16x16 C-tile.
+// Fragment maps hardware-verified by tools/probe_mfma.hip.
 template <int CI, int CO, int HW, int TS, bool RELU_IN, bool ADD_SKIP>
 __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
     const uint16_t* __restrict__ in,    // [P][CI][HW][HW]
@@ -23,70 +32,99 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
     const uint16_t* __restrict__ popw, int64_t row_stride, int64_t w_off,
     int64_t b_off, int P) {
   constexpr int TH = TS + 2;
+  constexpr int M = TS * TS;               // 256 or 64
+  constexpr int K = CI * 9;
+  constexpr int KSTEPS = (K + 31) / 32;
+  constexpr int NWAVES = M / 64;           // 4 or 1
+  constexpr int NTILES = (M / 16) * (CO / 16);
+  constexpr int MAXT = NTILES / NWAVES;    // tiles per wave
   __shared__ uint16_t a_lds[CI * TH * TH];
-  __shared__ uint16_t w_lds[CI * 9 * CO];
+  __shared__ uint16_t sa[M * 32];
+  __shared__ uint16_t sb[CO * 32];
   const int i = blockIdx.x;
   if (i >= P) return;
   constexpr int NT = HW / TS;
   const int ty0 = (blockIdx.y / NT) * TS, tx0 = (blockIdx.y % NT) * TS;
-  const int tid = threadIdx.x;
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
   const uint16_t* w = popw + (int64_t)i * row_stride;
 
-  // stage weights transposed [icc][co]
-  const uint16_t* wsrc = w + w_off;
-  for (int e = tid; e < CO * CI * 9; e += TS * TS) {
-    int co = e / (CI * 9), icc = e % (CI * 9);
-    w_lds[icc * CO + co] = wsrc[e];
-  }
-  // stage input halo tile (zero-padded at image edges)
+  // stage input halo tile (zero-padded at image edges, optional relu)
   const uint16_t* src = in + (int64_t)i * CI * HW * HW;
-  for (int e = tid; e < CI * TH * TH; e += TS * TS) {
+  for (int e = tid; e < CI * TH * TH; e += M) {
     int ci = e / (TH * TH), r = e % (TH * TH);
     int y = ty0 + r / TH - 1, x = tx0 + r % TH - 1;
     uint16_t v = 0;
     if (y >= 0 && y < HW && x >= 0 && x < HW) {
       v = src[(ci * HW + y) * HW + x];
-      if (RELU_IN && (v & 0x8000)) v = 0;  // relu on bf16: clear negatives
-    } else if (RELU_IN) {
-      v = 0;
+      if (RELU_IN && (v & 0x8000)) v = 0;
     }
     a_lds[e] = v;
   }
   __syncthreads();
 
-  const int py = tid / TS, px = tid % TS;          // position in tile
-  float acc[CO];
+  const uint16_t* wsrc = w + w_off;
+  mfma_cd acc[MAXT];
 #pragma unroll
-  for (int c = 0; c < CO; ++c) acc[c] = 0.f;
-  for (int ci = 0; ci < CI; ++ci) {
+  for (int u = 0; u < MAXT; ++u) acc[u] = 0.f;
+  for (int ks = 0; ks < KSTEPS; ++ks) {
+    // A-slab [M][32]: im2col gather from the halo tile
+    for (int e = tid; e < M * 32 / 4; e += M) {  // 4 elems per unit
+      int m = (e * 4) >> 5, kk0 = (e * 4) & 31;
+      int my = m / TS, mx = m % TS;
+      uint16_t vv[4];
 #pragma unroll
-    for (int dy = 0; dy < 3; ++dy) {
-#pragma unroll
-      for (int dx = 0; dx < 3; ++dx) {
-        float a = des::bf16_to_f32(
-            a_lds[(ci * TH + py + dy) * TH + px + dx]);
-        const uint16_t* wr = w_lds + ((ci * 9) + dy * 3 + dx) * CO;
-#pragma unroll
-        for (int c = 0; c < CO; c += 4) {
-          float4 w4 = des::load4(wr + c);
-          acc[c] += a * w4.x;
-          acc[c + 1] += a * w4.y;
-          acc[c + 2] += a * w4.z;
-          acc[c + 3] += a * w4.w;
+      for (int j = 0; j < 4; ++j) {
+        int k = ks * 32 + kk0 + j;
+        uint16_t v = 0;
+        if (k < K) {
+          int ci = k / 9, rm = k - ci * 9, ky = rm / 3, kx = rm - ky * 3;
+          v = a_lds[(ci * TH + my + ky) * TH + mx + kx];
         }
+        vv[j] = v;
       }
+      *reinterpret_cast<uint2*>(sa + m * 32 + kk0) =
+          uint2{(uint32_t)vv[0] | ((uint32_t)vv[1] << 16),
+                (uint32_t)vv[2] | ((uint32_t)vv[3] << 16)};
     }
+    // B-slab [CO][32]: direct rows of the flat [co][k] weights
+    for (int e = tid; e < CO * 32; e += M) {
+      int n = e >> 5, kk = e & 31;
+      int k = ks * 32 + kk;
+      sb[n * 32 + kk] = k < K ? wsrc[n * K + k] : (uint16_t)0;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int u = 0; u < MAXT; ++u) {
+      int t = wid + NWAVES * u;
+      int mt = t / (CO / 16), nt = t % (CO / 16);
+      mfma_ab af = *reinterpret_cast<const mfma_ab*>(
+          sa + (mt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
+      mfma_ab bf = *reinterpret_cast<const mfma_ab*>(
+          sb + (nt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
+      acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[u], 0, 0, 0);
+    }
+    __syncthreads();
   }
+
+  // epilogue: bias (+skip) -> global, D: lane l reg r -> (m, n)
   const uint16_t* bias = w + b_off;
   uint16_t* dst = out + (int64_t)i * CO * HW * HW;
-  const uint16_t* sk =
-      ADD_SKIP ? skip + (int64_t)i * CO * HW * HW : nullptr;
-  const int oy = ty0 + py, ox = tx0 + px;
+  const uint16_t* sk = ADD_SKIP ? skip + (int64_t)i * CO * HW * HW : nullptr;
 #pragma unroll
-  for (int c = 0; c < CO; ++c) {
-    float v = acc[c] + des::load1(bias + c);
-    if (ADD_SKIP) v += des::bf16_to_f32(sk[(c * HW + oy) * HW + ox]);
-    dst[(c * HW + oy) * HW + ox] = des::f32_to_bf16(v);
+  for (int u = 0; u < MAXT; ++u) {
+    int t = wid + NWAVES * u;
+    int mt = t / (CO / 16), nt = t % (CO / 16);
+    int n = nt * 16 + (lane & 15);
+    int m0 = mt * 16 + ((lane >> 4) << 2);
+    float b = des::bf16_to_f32(bias[n]);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + r;
+      int oy = ty0 + m / TS, ox = tx0 + m % TS;
+      float v = acc[u][r] + b;
+      if (ADD_SKIP) v += des::bf16_to_f32(sk[(n * HW + oy) * HW + ox]);
+      dst[(n * HW + oy) * HW + ox] = des::f32_to_bf16(v);
+    }
   }
 }
 
