@@ -41,6 +41,7 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
   __shared__ uint16_t a_lds[CI * TH * TH];
   __shared__ uint16_t sa[M * 32];
   __shared__ uint16_t sb[CO * 32];
+  __shared__ int tbl[32];  // k -> halo-tile offset (my=mx=0); -1 = zero-pad
   const int i = blockIdx.x;
   if (i >= P) return;
   constexpr int NT = HW / TS;
@@ -67,24 +68,29 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
 #pragma unroll
   for (int u = 0; u < MAXT; ++u) acc[u] = 0.f;
   for (int ks = 0; ks < KSTEPS; ++ks) {
+    // k -> offset table (separable: a_lds idx = tbl[kk] + my*TH + mx)
+    if (tid < 32) {
+      int k = ks * 32 + tid;
+      int off = -1;
+      if (k < K) {
+        int ci = k / 9, rm = k - ci * 9, ky = rm / 3, kx = rm - ky * 3;
+        off = (ci * TH + ky) * TH + kx;
+      }
+      tbl[tid] = off;
+    }
+    __syncthreads();
     // A-slab [M][32]: im2col gather from the halo tile
     for (int e = tid; e < M * 32 / 4; e += M) {  // 4 elems per unit
       int m = (e * 4) >> 5, kk0 = (e * 4) & 31;
-      int my = m / TS, mx = m % TS;
-      uint16_t vv[4];
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        int k = ks * 32 + kk0 + j;
-        uint16_t v = 0;
-        if (k < K) {
-          int ci = k / 9, rm = k - ci * 9, ky = rm / 3, kx = rm - ky * 3;
-          v = a_lds[(ci * TH + my + ky) * TH + mx + kx];
-        }
-        vv[j] = v;
-      }
+      int mybase = (m / TS) * TH + (m % TS);
+      const int4 tq = *reinterpret_cast<const int4*>(tbl + kk0);
+      uint16_t v0 = tq.x < 0 ? 0 : a_lds[tq.x + mybase];
+      uint16_t v1 = tq.y < 0 ? 0 : a_lds[tq.y + mybase];
+      uint16_t v2 = tq.z < 0 ? 0 : a_lds[tq.z + mybase];
+      uint16_t v3 = tq.w < 0 ? 0 : a_lds[tq.w + mybase];
       *reinterpret_cast<uint2*>(sa + m * 32 + kk0) =
-          uint2{(uint32_t)vv[0] | ((uint32_t)vv[1] << 16),
-                (uint32_t)vv[2] | ((uint32_t)vv[3] << 16)};
+          uint2{(uint32_t)v0 | ((uint32_t)v1 << 16),
+                (uint32_t)v2 | ((uint32_t)v3 << 16)};
     }
     // B-slab [CO][32]: direct rows of the flat [co][k] weights
     for (int e = tid; e < CO * 32; e += M) {
