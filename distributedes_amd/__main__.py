@@ -14,6 +14,9 @@ def main(argv=None):
     ap.add_argument("--set", nargs="*", default=[], metavar="K=V",
                     help="config overrides, e.g. --set pop=64 generations=10")
     ap.add_argument("--resume", default=None)
+    ap.add_argument("--evaluate", default=None, metavar="CKPT",
+                    help="evaluate a checkpoint (deterministic policy) and exit")
+    ap.add_argument("--episodes", type=int, default=16)
     ap.add_argument("--ops", default=None, choices=[None, "torch_ref", "hip"],
                     help="force ops backend (default: hip on GPU, torch on CPU)")
     args = ap.parse_args(argv)
@@ -26,6 +29,12 @@ def main(argv=None):
         except (ValueError, SyntaxError):
             overrides[k] = v
     cfg = preset(args.preset, **overrides)
+    if args.evaluate:
+        from .api import evaluate
+        stats = evaluate(args.evaluate, config=cfg, episodes=args.episodes,
+                         ops_backend=args.ops)
+        print(stats)
+        return stats
     state = run(cfg, resume=args.resume, ops_backend=args.ops)
     return state
 

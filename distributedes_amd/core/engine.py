@@ -99,8 +99,10 @@ class TrainState:
         want_hip = (ops_backend != "torch_ref")
         self.ops = _make_ops(self.device, want_hip) if ops_backend != "torch_ref" \
             else _make_ops(torch.device("cpu"), False)
+        table_dtype = (torch.bfloat16 if cfg.noise_dtype == "bf16"
+                       else torch.float32)
         self.table = self.ops.fill_table(cfg.noise_table_size, self.table_seed,
-                                         self.device)
+                                         self.device, dtype=table_dtype)
 
         self.generation = 0
         self.env_steps_total = 0

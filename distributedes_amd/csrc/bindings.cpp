@@ -10,13 +10,11 @@
 
 extern "C" {
 void des_launch_noise_fill(float*, int64_t, uint64_t, hipStream_t);
-void des_launch_perturb_f32(float*, const float*, const float*, const int64_t*,
-                            const float*, float, const float*, int64_t,
-                            int64_t, int64_t, hipStream_t);
-void des_launch_perturb_bf16(uint16_t*, const float*, const float*,
-                             const int64_t*, const float*, float, const float*,
-                             int64_t, int64_t, int64_t, hipStream_t);
-void des_launch_wsum(float*, float*, const float*, const int64_t*,
+void des_launch_noise_fill_bf16(uint16_t*, int64_t, uint64_t, hipStream_t);
+void des_launch_perturb(void*, bool, const float*, const void*, bool,
+                        const int64_t*, const float*, float, const float*,
+                        int64_t, int64_t, int64_t, hipStream_t);
+void des_launch_wsum(float*, float*, const void*, bool, const int64_t*,
                      const float*, const float*, int64_t, int64_t, bool,
                      hipStream_t);
 void des_launch_mlp_forward(const void*, bool, int64_t, const float*, int,
@@ -78,9 +76,15 @@ uint8_t* bptr(torch::Tensor& t) { return t.data_ptr<uint8_t>(); }
 
 void noise_fill(torch::Tensor table, int64_t seed) {
   CHECK_DEV(table); CHECK_CONT(table);
-  TORCH_CHECK(table.dtype() == torch::kFloat32, "table must be fp32");
-  des_launch_noise_fill(table.data_ptr<float>(), table.numel(),
-                        (uint64_t)seed, cur_stream());
+  if (table.dtype() == torch::kFloat32)
+    des_launch_noise_fill(table.data_ptr<float>(), table.numel(),
+                          (uint64_t)seed, cur_stream());
+  else if (table.dtype() == torch::kBFloat16)
+    des_launch_noise_fill_bf16(
+        reinterpret_cast<uint16_t*>(table.data_ptr()), table.numel(),
+        (uint64_t)seed, cur_stream());
+  else
+    TORCH_CHECK(false, "table must be fp32 or bf16");
 }
 
 void perturb(torch::Tensor out, torch::Tensor theta, torch::Tensor table,
@@ -98,19 +102,16 @@ void perturb(torch::Tensor out, torch::Tensor theta, torch::Tensor table,
     TORCH_CHECK(sigma_vec.numel() == D, "sigma_vec must be [D]");
     sv = sigma_vec.data_ptr<float>();
   }
-  if (out.dtype() == torch::kBFloat16) {
-    des_launch_perturb_bf16(
-        reinterpret_cast<uint16_t*>(out.data_ptr()), theta.data_ptr<float>(),
-        table.data_ptr<float>(), offsets.data_ptr<int64_t>(),
-        signs.data_ptr<float>(), (float)sigma, sv, P, D, stride, cur_stream());
-  } else if (out.dtype() == torch::kFloat32) {
-    des_launch_perturb_f32(out.data_ptr<float>(), theta.data_ptr<float>(),
-                           table.data_ptr<float>(),
-                           offsets.data_ptr<int64_t>(), signs.data_ptr<float>(),
-                           (float)sigma, sv, P, D, stride, cur_stream());
-  } else {
-    TORCH_CHECK(false, "perturb: out must be bf16 or fp32");
-  }
+  bool out_bf16 = out.dtype() == torch::kBFloat16;
+  bool tab_bf16 = table.dtype() == torch::kBFloat16;
+  TORCH_CHECK(out_bf16 || out.dtype() == torch::kFloat32,
+              "perturb: out must be bf16 or fp32");
+  TORCH_CHECK(tab_bf16 || table.dtype() == torch::kFloat32,
+              "perturb: table must be bf16 or fp32");
+  des_launch_perturb(out.data_ptr(), out_bf16, theta.data_ptr<float>(),
+                     table.data_ptr(), tab_bf16, offsets.data_ptr<int64_t>(),
+                     signs.data_ptr<float>(), (float)sigma, sv, P, D, stride,
+                     cur_stream());
 }
 
 void weighted_noise_sum(torch::Tensor acc, torch::Tensor acc_sq,
@@ -123,9 +124,9 @@ void weighted_noise_sum(torch::Tensor acc, torch::Tensor acc_sq,
   if (want_sq) TORCH_CHECK(acc_sq.numel() == D);
   des_launch_wsum(acc.data_ptr<float>(),
                   want_sq ? acc_sq.data_ptr<float>() : nullptr,
-                  table.data_ptr<float>(), offsets.data_ptr<int64_t>(),
-                  signs.data_ptr<float>(), weights.data_ptr<float>(), P, D,
-                  want_sq, cur_stream());
+                  table.data_ptr(), table.dtype() == torch::kBFloat16,
+                  offsets.data_ptr<int64_t>(), signs.data_ptr<float>(),
+                  weights.data_ptr<float>(), P, D, want_sq, cur_stream());
 }
 
 // ---------------- policy forwards ----------------

@@ -9,16 +9,17 @@
 #include <algorithm>
 
 #include "common.h"
+#include "gemv.h"
 
 namespace {
 
 // --------------------------------------------------------------------------
 // perturb: 2D grid (chunks, members).  Each thread writes 4 elements.
 // sigma_vec == nullptr -> scalar sigma (ES); else per-coordinate (CEM).
-template <typename OutT>
+template <typename OutT, typename TT>
 __global__ void perturb_kernel(OutT* __restrict__ out,
                                const float* __restrict__ theta,
-                               const float* __restrict__ table,
+                               const TT* __restrict__ table,
                                const int64_t* __restrict__ offsets,
                                const float* __restrict__ signs, float sigma,
                                const float* __restrict__ sigma_vec, int64_t P,
@@ -26,7 +27,7 @@ __global__ void perturb_kernel(OutT* __restrict__ out,
   int64_t i = blockIdx.y;  // member
   if (i >= P) return;
   const float s = signs[i] * (sigma_vec ? 1.0f : sigma);
-  const float* src = table + offsets[i];
+  const TT* src = table + offsets[i];
   OutT* dst = out + i * out_stride;
   // rows are normally padded to 16B (out_stride % 8 == 0) so vector stores
   // stay aligned; unpadded odd-D rows fall back to element stores
@@ -36,7 +37,7 @@ __global__ void perturb_kernel(OutT* __restrict__ out,
   for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; v < d4;
        v += stride) {
     int64_t d = v << 2;
-    float4 eps = *reinterpret_cast<const float4*>(src + d);
+    float4 eps = des::load4(src + d);
     float4 th = *reinterpret_cast<const float4*>(theta + d);
     float4 sv = sigma_vec ? *reinterpret_cast<const float4*>(sigma_vec + d)
                           : float4{1.f, 1.f, 1.f, 1.f};
@@ -66,7 +67,7 @@ __global__ void perturb_kernel(OutT* __restrict__ out,
   if (blockIdx.x == 0 && threadIdx.x == 0) {
     for (int64_t d = d4 << 2; d < D; ++d) {
       float sv = sigma_vec ? sigma_vec[d] : 1.0f;
-      float r = theta[d] + s * sv * src[d];
+      float r = theta[d] + s * sv * des::load1(src + d);
       if constexpr (sizeof(OutT) == 2)
         reinterpret_cast<uint16_t*>(dst)[d] = des::f32_to_bf16(r);
       else
@@ -80,9 +81,9 @@ __global__ void perturb_kernel(OutT* __restrict__ out,
 // loops over all P members, streaming their noise slices (coalesced float4).
 // Members with w_i == 0 are skipped (wave-uniform branch -> CEM elite subset
 // costs only elite traffic).  fp32 accumulate in registers.
-template <bool WANT_SQ>
+template <bool WANT_SQ, typename TT>
 __global__ void wsum_kernel(float* __restrict__ acc, float* __restrict__ acc_sq,
-                            const float* __restrict__ table,
+                            const TT* __restrict__ table,
                             const int64_t* __restrict__ offsets,
                             const float* __restrict__ signs,
                             const float* __restrict__ weights, int64_t P,
@@ -97,7 +98,7 @@ __global__ void wsum_kernel(float* __restrict__ acc, float* __restrict__ acc_sq,
       float w = weights[i];
       if (w == 0.0f) continue;
       float ws = w * signs[i];
-      float4 eps = *reinterpret_cast<const float4*>(table + offsets[i] + d);
+      float4 eps = des::load4(table + offsets[i] + d);
       a.x += ws * eps.x; a.y += ws * eps.y;
       a.z += ws * eps.z; a.w += ws * eps.w;
       if constexpr (WANT_SQ) {
@@ -115,7 +116,7 @@ __global__ void wsum_kernel(float* __restrict__ acc, float* __restrict__ acc_sq,
       for (int64_t i = 0; i < P; ++i) {
         float w = weights[i];
         if (w == 0.0f) continue;
-        float eps = table[offsets[i] + d];
+        float eps = des::load1(table + offsets[i] + d);
         a += w * signs[i] * eps;
         if constexpr (WANT_SQ) q += w * eps * eps;
       }
@@ -127,45 +128,65 @@ __global__ void wsum_kernel(float* __restrict__ acc, float* __restrict__ acc_sq,
 
 }  // namespace
 
-extern "C" {
-
-void des_launch_perturb_f32(float* out, const float* theta, const float* table,
-                            const int64_t* offsets, const float* signs,
-                            float sigma, const float* sigma_vec, int64_t P,
-                            int64_t D, int64_t out_stride, hipStream_t stream) {
+template <typename OutT, typename TT>
+static void launch_perturb(OutT* out, const float* theta, const TT* table,
+                           const int64_t* offsets, const float* signs,
+                           float sigma, const float* sigma_vec, int64_t P,
+                           int64_t D, int64_t out_stride, hipStream_t stream) {
   int threads = 256;
   int chunks = (int)std::min<int64_t>(((D >> 2) + threads - 1) / threads, 64);
-  hipLaunchKernelGGL(perturb_kernel<float>, dim3(chunks, (uint32_t)P),
+  hipLaunchKernelGGL((perturb_kernel<OutT, TT>), dim3(chunks, (uint32_t)P),
                      dim3(threads), 0, stream, out, theta, table, offsets,
                      signs, sigma, sigma_vec, P, D, out_stride);
 }
 
-void des_launch_perturb_bf16(uint16_t* out, const float* theta,
-                             const float* table, const int64_t* offsets,
-                             const float* signs, float sigma,
-                             const float* sigma_vec, int64_t P, int64_t D,
-                             int64_t out_stride, hipStream_t stream) {
-  int threads = 256;
-  int chunks = (int)std::min<int64_t>(((D >> 2) + threads - 1) / threads, 64);
-  hipLaunchKernelGGL(perturb_kernel<uint16_t>, dim3(chunks, (uint32_t)P),
-                     dim3(threads), 0, stream, out, theta, table, offsets,
-                     signs, sigma, sigma_vec, P, D, out_stride);
+extern "C" void des_launch_perturb(void* out, bool out_bf16, const float* theta,
+                        const void* table, bool table_bf16,
+                        const int64_t* offsets, const float* signs,
+                        float sigma, const float* sigma_vec, int64_t P,
+                        int64_t D, int64_t out_stride, hipStream_t stream) {
+  if (out_bf16 && table_bf16)
+    launch_perturb((uint16_t*)out, theta, (const uint16_t*)table, offsets,
+                   signs, sigma, sigma_vec, P, D, out_stride, stream);
+  else if (out_bf16)
+    launch_perturb((uint16_t*)out, theta, (const float*)table, offsets, signs,
+                   sigma, sigma_vec, P, D, out_stride, stream);
+  else if (table_bf16)
+    launch_perturb((float*)out, theta, (const uint16_t*)table, offsets, signs,
+                   sigma, sigma_vec, P, D, out_stride, stream);
+  else
+    launch_perturb((float*)out, theta, (const float*)table, offsets, signs,
+                   sigma, sigma_vec, P, D, out_stride, stream);
 }
 
-void des_launch_wsum(float* acc, float* acc_sq, const float* table,
-                     const int64_t* offsets, const float* signs,
-                     const float* weights, int64_t P, int64_t D, bool want_sq,
-                     hipStream_t stream) {
+template <typename TT>
+static void launch_wsum(float* acc, float* acc_sq, const TT* table,
+                        const int64_t* offsets, const float* signs,
+                        const float* weights, int64_t P, int64_t D,
+                        bool want_sq, hipStream_t stream) {
   int threads = 256;
   int64_t d4 = D >> 2;
   int grid = (int)std::max<int64_t>((d4 + threads - 1) / threads, 1);
   if (want_sq)
-    hipLaunchKernelGGL(wsum_kernel<true>, dim3(grid), dim3(threads), 0, stream,
-                       acc, acc_sq, table, offsets, signs, weights, P, D);
+    hipLaunchKernelGGL((wsum_kernel<true, TT>), dim3(grid), dim3(threads), 0,
+                       stream, acc, acc_sq, table, offsets, signs, weights, P,
+                       D);
   else
-    hipLaunchKernelGGL(wsum_kernel<false>, dim3(grid), dim3(threads), 0,
+    hipLaunchKernelGGL((wsum_kernel<false, TT>), dim3(grid), dim3(threads), 0,
                        stream, acc, acc_sq, table, offsets, signs, weights, P,
                        D);
 }
 
-}  // extern "C"
+extern "C" void des_launch_wsum(float* acc, float* acc_sq, const void* table,
+                     bool table_bf16, const int64_t* offsets,
+                     const float* signs, const float* weights, int64_t P,
+                     int64_t D, bool want_sq, hipStream_t stream) {
+  if (table_bf16)
+    launch_wsum(acc, acc_sq, (const uint16_t*)table, offsets, signs, weights,
+                P, D, want_sq, stream);
+  else
+    launch_wsum(acc, acc_sq, (const float*)table, offsets, signs, weights, P,
+                D, want_sq, stream);
+}
+
+

@@ -51,8 +51,9 @@ class HipOps:
             raise RuntimeError(f"HIP extension not built: {_err}")
 
     # K1
-    def fill_table(self, size: int, seed: int, device) -> torch.Tensor:
-        t = torch.empty(size, dtype=torch.float32, device=device)
+    def fill_table(self, size: int, seed: int, device,
+                   dtype=torch.float32) -> torch.Tensor:
+        t = torch.empty(size, dtype=dtype, device=device)
         self.ext.noise_fill(t, seed)
         return t
 

@@ -16,9 +16,10 @@ class TorchOps:
     name = "torch_ref"
 
     # --- K1 ---------------------------------------------------------------
-    def fill_table(self, size: int, seed: int, device) -> torch.Tensor:
+    def fill_table(self, size: int, seed: int, device,
+                   dtype=torch.float32) -> torch.Tensor:
         t = noise_mod.fill_noise_table_cpu(size, seed)
-        return t.to(device)
+        return t.to(dtype).to(device)
 
     # --- K2: theta_i = theta + sigma * sign_i * table[off_i : off_i+D] ----
     def perturb(self, theta: torch.Tensor, table: torch.Tensor,

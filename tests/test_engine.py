@@ -94,3 +94,17 @@ def test_sigterm_checkpoint(tmp_path):
     assert os.path.exists(p)
     ck = load_checkpoint(p)
     assert ck["generation"] == 2
+
+
+def test_bf16_noise_table_end_to_end():
+    """cfg.noise_dtype='bf16': table stored bf16 (half the memory/traffic),
+    training still works and stays deterministic."""
+    cfg = _mini("es").replace(noise_dtype="bf16")
+    a = TrainState(cfg, ops_backend="torch_ref")
+    b = TrainState(cfg, ops_backend="torch_ref")
+    assert a.table.dtype == torch.bfloat16
+    for _ in range(2):
+        train_step(a)
+        train_step(b)
+    assert torch.equal(a.theta, b.theta)
+    assert torch.isfinite(a.theta).all()

@@ -97,3 +97,23 @@ def test_native_extension_is_loaded_on_gpu():
     from distributedes_amd.core.engine import _make_ops
     ops = _make_ops(torch.device("cuda"))
     assert ops.name == "hip"
+
+
+@pytest.mark.parametrize("table_dtype", [torch.bfloat16])
+def test_bf16_table_fill_and_perturb(ops, ref, table_dtype):
+    n = 1 << 14
+    gpu = ops.fill_table(n, 4242, "cuda", dtype=table_dtype)
+    cpu = ref.fill_table(n, 4242, "cpu", dtype=table_dtype)
+    # same Philox bits + same bf16 rounding -> near-exact
+    torch.testing.assert_close(gpu.cpu().float(), cpu.float(),
+                               rtol=1e-2, atol=1e-2)
+    D = 1024
+    theta = torch.randn(D, device="cuda")
+    offsets = (torch.randint(0, (n - D) // 16, (8,), device="cuda") * 16).to(torch.int64)
+    signs = torch.ones(8, device="cuda")
+    got = ops.perturb(theta, gpu, offsets, signs, 0.1)
+    want = ref.perturb(theta.cpu(), cpu, offsets.cpu(), signs.cpu(), 0.1)
+    torch.testing.assert_close(got.cpu(), want, rtol=1e-4, atol=1e-4)
+    acc = ops.weighted_noise_sum(gpu, offsets, signs,
+                                 torch.rand(8, device="cuda"), D)
+    assert torch.isfinite(acc).all()
