@@ -103,4 +103,55 @@ __device__ void gemv_rows4(const WT* __restrict__ w, const WT* __restrict__ b,
   }
 }
 
+// bf16-weight variant with 16-B (8-element) loads: preferred whenever the
+// layer offset and row stride are 8-element aligned (checked by caller)
+template <int ACT>
+__device__ void gemv_rows4_w8(const uint16_t* __restrict__ w,
+                              const uint16_t* __restrict__ b,
+                              const float* __restrict__ x,
+                              float* __restrict__ y, int in_dim, int out_dim,
+                              int64_t stride, int lane, int wid, int nwaves) {
+  int full = in_dim & ~7;
+  for (int r0 = wid * 4; r0 < out_dim; r0 += nwaves * 4) {
+    int rc = out_dim - 1;
+    const uint16_t* w0 = w + (int64_t)r0 * stride;
+    const uint16_t* w1 = w + (int64_t)min(r0 + 1, rc) * stride;
+    const uint16_t* w2 = w + (int64_t)min(r0 + 2, rc) * stride;
+    const uint16_t* w3 = w + (int64_t)min(r0 + 3, rc) * stride;
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    for (int i = lane * 8; i < full; i += 64 * 8) {
+      uint4 v0 = *reinterpret_cast<const uint4*>(w0 + i);
+      uint4 v1 = *reinterpret_cast<const uint4*>(w1 + i);
+      uint4 v2 = *reinterpret_cast<const uint4*>(w2 + i);
+      uint4 v3 = *reinterpret_cast<const uint4*>(w3 + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xv = x[i + j];
+        a0 += xv * bf16_to_f32(reinterpret_cast<const uint16_t*>(&v0)[j]);
+        a1 += xv * bf16_to_f32(reinterpret_cast<const uint16_t*>(&v1)[j]);
+        a2 += xv * bf16_to_f32(reinterpret_cast<const uint16_t*>(&v2)[j]);
+        a3 += xv * bf16_to_f32(reinterpret_cast<const uint16_t*>(&v3)[j]);
+      }
+    }
+    for (int i = full + lane; i < in_dim; i += 64) {
+      float xv = x[i];
+      a0 += load1(w0 + i) * xv;
+      a1 += load1(w1 + i) * xv;
+      a2 += load1(w2 + i) * xv;
+      a3 += load1(w3 + i) * xv;
+    }
+    a0 = wave_reduce_sum_f(a0);
+    a1 = wave_reduce_sum_f(a1);
+    a2 = wave_reduce_sum_f(a2);
+    a3 = wave_reduce_sum_f(a3);
+    if (lane < 4 && r0 + lane < out_dim) {
+      float acc = lane == 0 ? a0 : lane == 1 ? a1 : lane == 2 ? a2 : a3;
+      acc += load1(b + r0 + lane);
+      if (ACT == 1) acc = tanhf(acc);
+      if (ACT == 2) acc = fmaxf(acc, 0.f);
+      y[r0 + lane] = acc;
+    }
+  }
+}
+
 }  // namespace des

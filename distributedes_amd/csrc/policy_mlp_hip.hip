@@ -45,20 +45,37 @@ __global__ __launch_bounds__(256) void mlp_pop_forward_kernel(
   }
   __syncthreads();
 
-  des::gemv_rows4<WT, 1>(w + ow1, w + ob1, xbuf, ybuf, in_dim, h1, in_dim,
-                         lane, wid, 4);
+  // bf16 weights with 8-elem-aligned layer offset + row stride -> 16-B loads
+  const bool w8_1 = sizeof(WT) == 2 && (ow1 & 7) == 0 && (in_dim & 7) == 0;
+  if (w8_1)
+    des::gemv_rows4_w8<1>((const uint16_t*)(w + ow1),
+                          (const uint16_t*)(w + ob1), xbuf, ybuf, in_dim, h1,
+                          in_dim, lane, wid, 4);
+  else
+    des::gemv_rows4<WT, 1>(w + ow1, w + ob1, xbuf, ybuf, in_dim, h1, in_dim,
+                           lane, wid, 4);
   __syncthreads();
   const float* x2 = ybuf;
   float* y2 = xbuf;
   int last_dim = h1;
   if (h2 > 0) {
-    des::gemv_rows4<WT, 1>(w + ow2, w + ob2, ybuf, xbuf, h1, h2, h1, lane,
-                           wid, 4);
+    if (sizeof(WT) == 2 && (ow2 & 7) == 0 && (h1 & 7) == 0)
+      des::gemv_rows4_w8<1>((const uint16_t*)(w + ow2),
+                            (const uint16_t*)(w + ob2), ybuf, xbuf, h1, h2,
+                            h1, lane, wid, 4);
+    else
+      des::gemv_rows4<WT, 1>(w + ow2, w + ob2, ybuf, xbuf, h1, h2, h1, lane,
+                             wid, 4);
     __syncthreads();
     x2 = xbuf; y2 = ybuf; last_dim = h2;
   }
-  des::gemv_rows4<WT, 0>(w + owh, w + obh, x2, y2, last_dim, A, last_dim,
-                         lane, wid, 4);
+  if (sizeof(WT) == 2 && (owh & 7) == 0 && (last_dim & 7) == 0)
+    des::gemv_rows4_w8<0>((const uint16_t*)(w + owh),
+                          (const uint16_t*)(w + obh), x2, y2, last_dim, A,
+                          last_dim, lane, wid, 4);
+  else
+    des::gemv_rows4<WT, 0>(w + owh, w + obh, x2, y2, last_dim, A, last_dim,
+                           lane, wid, 4);
   __syncthreads();
 
   if (logits_out)
