@@ -102,7 +102,6 @@ def main():
     # MAX over ranks (slowest rank defines job time)
     el = torch.tensor([elapsed], device=state.device
                       if state.comm.backend == "nccl" else "cpu")
-    del s
     if state.comm.is_dist:
         import torch.distributed as dist
         dist.all_reduce(el, op=dist.ReduceOp.MAX)
