@@ -108,3 +108,20 @@ def test_bf16_noise_table_end_to_end():
         train_step(b)
     assert torch.equal(a.theta, b.theta)
     assert torch.isfinite(a.theta).all()
+
+
+def test_metrics_jsonl_and_plot(tmp_path):
+    """R8 parity: per-generation JSONL metrics -> SVG learning curve."""
+    mp = str(tmp_path / "m.jsonl")
+    cfg = _mini("es", metrics_path=mp)
+    st = TrainState(cfg, ops_backend="torch_ref")
+    for _ in range(3):
+        train_step(st)
+    st.metrics.close()
+    from distributedes_amd.utils.plot import load_series, render_svg
+    xs, ys = load_series(mp)
+    assert xs == [1, 2, 3]
+    svg_path = str(tmp_path / "c.svg")
+    svg = render_svg({"ret": (xs, ys)})
+    open(svg_path, "w").write(svg)
+    assert "<svg" in svg and "polyline" in svg
