@@ -67,7 +67,7 @@ def evaluate(checkpoint_or_state, config: Optional[Config] = None,
         state = checkpoint_or_state
     else:
         assert config is not None, "evaluate(path) needs the config"
-        cfg = config.replace(pop=episodes * max(1, 1))
+        cfg = config.replace(pop=episodes if episodes % 2 == 0 else episodes + 1)
         state = TrainState(cfg, ops_backend=ops_backend)
         state.load_payload(load_checkpoint(checkpoint_or_state))
 
