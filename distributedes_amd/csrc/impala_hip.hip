@@ -21,12 +21,10 @@ typedef __attribute__((ext_vector_type(4))) float mfma_cd;
 
 // 3x3 pad-1 conv as MFMA implicit GEMM per spatial tile:
 //   M = TS*TS positions, N = CO, K = CI*9 (zero-padded to 32-multiples).
-// The member's whole [CO][K] weight tile stages ONCE into LDS (coalesced
-// burst, row-padded so direct b128 B-fragment reads are bank-spread) — the
-// per-k-step loop then touches only LDS, so the member-unique weight HBM
-// latency is paid once per block instead of once per k-step.  All 32-entry
-// im2col offset tables are precomputed up front.  Fragment maps verified by
-// tools/probe_mfma.hip.
+// Per K-step all threads build the A-slab (im2col from the halo-staged
+// input tile) and B-slab (direct [co][k] rows of the member's weights),
+// then each wave runs one v_mfma_f32_16x16x32_bf16 per owned 16x16 C-tile.
+// Fragment maps hardware-verified by tools/probe_mfma.hip.
 template <int CI, int CO, int HW, int TS, bool RELU_IN, bool ADD_SKIP>
 __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
     const uint16_t* __restrict__ in,    // [P][CI][HW][HW]
@@ -38,14 +36,13 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
   constexpr int M = TS * TS;               // 256 or 64
   constexpr int K = CI * 9;
   constexpr int KSTEPS = (K + 31) / 32;
-  constexpr int KPAD = KSTEPS * 32 + 8;    // covers every k-step read; 8-aligned pad spreads banks
   constexpr int NWAVES = M / 64;           // 4 or 1
   constexpr int NTILES = (M / 16) * (CO / 16);
   constexpr int MAXT = NTILES / NWAVES;    // tiles per wave
   __shared__ uint16_t a_lds[CI * TH * TH];
   __shared__ uint16_t sa[M * 32];
-  __shared__ uint16_t wt[CO * KPAD];
-  __shared__ int tbl[KSTEPS * 32];
+  __shared__ uint16_t sb[CO * 32];
+  __shared__ int tbl[32];  // k -> halo-tile offset (my=mx=0); -1 = zero-pad
   const int i = blockIdx.x;
   if (i >= P) return;
   constexpr int NT = HW / TS;
@@ -53,7 +50,7 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
   const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
   const uint16_t* w = popw + (int64_t)i * row_stride;
 
-  // halo tile (zero-padded at image edges, optional fused relu)
+  // stage input halo tile (zero-padded at image edges, optional relu)
   const uint16_t* src = in + (int64_t)i * CI * HW * HW;
   for (int e = tid; e < CI * TH * TH; e += M) {
     int ci = e / (TH * TH), r = e % (TH * TH);
@@ -65,32 +62,29 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
     }
     a_lds[e] = v;
   }
-  // full weight tile, one coalesced burst (k zero-padded to KPAD)
-  const uint16_t* wsrc = w + w_off;
-  for (int e = tid; e < CO * KPAD; e += M) {
-    int n = e / KPAD, k = e - n * KPAD;
-    wt[e] = k < K ? wsrc[n * K + k] : (uint16_t)0;
-  }
-  // im2col offset tables (k -> halo offset at my = mx = 0; -1 = pad)
-  for (int e = tid; e < KSTEPS * 32; e += M) {
-    int off = -1;
-    if (e < K) {
-      int ci = e / 9, rm = e - ci * 9, ky = rm / 3, kx = rm - ky * 3;
-      off = (ci * TH + ky) * TH + kx;
-    }
-    tbl[e] = off;
-  }
   __syncthreads();
 
+  const uint16_t* wsrc = w + w_off;
   mfma_cd acc[MAXT];
 #pragma unroll
   for (int u = 0; u < MAXT; ++u) acc[u] = 0.f;
   for (int ks = 0; ks < KSTEPS; ++ks) {
+    // k -> offset table (separable: a_lds idx = tbl[kk] + my*TH + mx)
+    if (tid < 32) {
+      int k = ks * 32 + tid;
+      int off = -1;
+      if (k < K) {
+        int ci = k / 9, rm = k - ci * 9, ky = rm / 3, kx = rm - ky * 3;
+        off = (ci * TH + ky) * TH + kx;
+      }
+      tbl[tid] = off;
+    }
+    __syncthreads();
     // A-slab [M][32]: im2col gather from the halo tile
-    for (int e = tid; e < M * 8; e += M) {   // 4 elems per unit
-      int m = e >> 3, kk0 = (e & 7) * 4;
+    for (int e = tid; e < M * 32 / 4; e += M) {  // 4 elems per unit
+      int m = (e * 4) >> 5, kk0 = (e * 4) & 31;
       int mybase = (m / TS) * TH + (m % TS);
-      const int4 tq = *reinterpret_cast<const int4*>(tbl + ks * 32 + kk0);
+      const int4 tq = *reinterpret_cast<const int4*>(tbl + kk0);
       uint16_t v0 = tq.x < 0 ? 0 : a_lds[tq.x + mybase];
       uint16_t v1 = tq.y < 0 ? 0 : a_lds[tq.y + mybase];
       uint16_t v2 = tq.z < 0 ? 0 : a_lds[tq.z + mybase];
@@ -98,6 +92,12 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
       *reinterpret_cast<uint2*>(sa + m * 32 + kk0) =
           uint2{(uint32_t)v0 | ((uint32_t)v1 << 16),
                 (uint32_t)v2 | ((uint32_t)v3 << 16)};
+    }
+    // B-slab [CO][32]: direct rows of the flat [co][k] weights
+    for (int e = tid; e < CO * 32; e += M) {
+      int n = e >> 5, kk = e & 31;
+      int k = ks * 32 + kk;
+      sb[n * 32 + kk] = k < K ? wsrc[n * K + k] : (uint16_t)0;
     }
     __syncthreads();
 #pragma unroll
@@ -107,7 +107,7 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
       mfma_ab af = *reinterpret_cast<const mfma_ab*>(
           sa + (mt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
       mfma_ab bf = *reinterpret_cast<const mfma_ab*>(
-          wt + (nt * 16 + (lane & 15)) * KPAD + ks * 32 + (lane >> 4) * 8);
+          sb + (nt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
       acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[u], 0, 0, 0);
     }
     __syncthreads();
