@@ -67,30 +67,7 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
   mfma_cd acc[MAXT];
 #pragma unroll
   for (int u = 0; u < MAXT; ++u) acc[u] = 0.f;
-  // B-slab for ks = 0 (member-unique weights: full HBM latency, paid here
-  // once; later steps prefetch into registers under the current MFMA)
-  constexpr int BLD = (CO * 32 + M - 1) / M;   // B elements per thread
-  for (int e = tid; e < CO * 32; e += M) {
-    int n = e >> 5, kk = e & 31;
-    int k = kk;
-    sb[n * 32 + kk] = k < K ? wsrc[n * K + k] : (uint16_t)0;
-  }
   for (int ks = 0; ks < KSTEPS; ++ks) {
-    // prefetch next k-step's B into registers (latency hides under the
-    // A-gather + MFMA below; stored to LDS after the post-MFMA barrier)
-    uint16_t bpre[BLD];
-    if (ks + 1 < KSTEPS) {
-#pragma unroll
-      for (int j = 0; j < BLD; ++j) {
-        int e = tid + j * M;
-        uint16_t v = 0;
-        if (e < CO * 32) {
-          int n = e >> 5, k = (ks + 1) * 32 + (e & 31);
-          v = k < K ? wsrc[n * K + k] : (uint16_t)0;
-        }
-        bpre[j] = v;
-      }
-    }
     // k -> offset table (separable: a_lds idx = tbl[kk] + my*TH + mx)
     if (tid < 32) {
       int k = ks * 32 + tid;
@@ -133,15 +110,6 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
       acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[u], 0, 0, 0);
     }
     __syncthreads();
-    if (ks + 1 < KSTEPS) {
-#pragma unroll
-      for (int j = 0; j < BLD; ++j) {
-        int e = tid + j * M;
-        if (e < CO * 32) sb[e] = bpre[j];
-      }
-    }
-    // (next iteration's post-gather barrier orders these stores before the
-    // MFMA that consumes them)
   }
 
   // epilogue: bias (+skip) -> global, D: lane l reg r -> (m, n)
