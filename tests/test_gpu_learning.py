@@ -55,3 +55,19 @@ def test_pong_es_improves_on_gpu():
         best = max(best, train_step(st)["episodic_return_mean"])
     assert best > first - 0.5  # not diverging; usually improves
     assert torch.isfinite(st.theta).all()
+
+
+def test_pong_throughput_floor():
+    """SURVEY §4.2 integration tier: short Pong-shaped throughput smoke with a
+    floor assertion (generous: ~half the measured 1.66M steps/s)."""
+    import time
+    cfg = preset("pong_es_8k", pop=8192, episode_steps=120,
+                 noise_table_size=1 << 24, log_every=10 ** 9)
+    st = TrainState(cfg)
+    train_step(st)                       # warmup (graph capture)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    s = train_step(st)
+    torch.cuda.synchronize()
+    rate = s["env_steps"] / (time.perf_counter() - t0)
+    assert rate > 800_000, rate
