@@ -6,7 +6,7 @@ matching the five BASELINE configs exactly.
 from __future__ import annotations
 
 import dataclasses
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 

@@ -24,7 +24,6 @@ import math
 import time
 from typing import Optional
 
-import numpy as np
 import torch
 
 from ..config import Config
@@ -32,7 +31,7 @@ from ..envs import make_env
 from ..models import make_model
 from ..parallel import Comm
 from ..utils import (IdentityNormalizer, MetricsLogger, RunningNormalizer,
-                     load_checkpoint, save_checkpoint)
+                     save_checkpoint)
 from . import noise as noise_mod
 from .optim import make_optimizer
 from .rank import shape_fitness

@@ -8,7 +8,6 @@ eager fallback (the engine enforces this in _make_ops).
 from __future__ import annotations
 
 import importlib
-import os
 
 import torch
 
