@@ -20,7 +20,8 @@ ROOT = Path(__file__).parent
 CSRC = ROOT / "distributedes_amd" / "csrc"
 
 sources = sorted(str(p) for p in CSRC.glob("*.cpp")) + \
-          sorted(str(p) for p in CSRC.glob("*.hip"))
+          sorted(str(p) for p in CSRC.glob("*.hip")
+                 if not p.name.endswith("_hip.hip"))  # exclude hipify copies
 
 setup(
     name="distributedes_amd_hip",

@@ -125,3 +125,19 @@ def test_metrics_jsonl_and_plot(tmp_path):
     svg = render_svg({"ret": (xs, ys)})
     open(svg_path, "w").write(svg)
     assert "<svg" in svg and "polyline" in svg
+
+
+def test_rank_shaping_modes_run():
+    for mode in ("normalized", "raw"):
+        st = TrainState(_mini("es", rank_shaping=mode), ops_backend="torch_ref")
+        t0 = st.theta.clone()
+        train_step(st)
+        assert not torch.equal(st.theta, t0)
+        assert torch.isfinite(st.theta).all()
+
+
+def test_sgd_optimizer_runs():
+    st = TrainState(_mini("es", optimizer="sgd"), ops_backend="torch_ref")
+    train_step(st)
+    assert st.optimizer.t == 1
+    assert torch.isfinite(st.theta).all()
