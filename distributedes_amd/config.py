@@ -65,6 +65,8 @@ class Config:
 
     # --- misc ---
     theta_allreduce_every: int = 10  # drift-insurance parameter all-reduce cadence
+    profile_phases: bool = False     # per-phase timings in train_step stats
+                                     # (adds device syncs)
 
     def replace(self, **kw) -> "Config":
         return dataclasses.replace(self, **kw)

@@ -275,9 +275,17 @@ def train_step(state: TrainState) -> dict:
     """Exactly one generation.  Returns the stats dict."""
     cfg = state.cfg
     t0 = time.perf_counter()
+
+    def _phase_mark():
+        if cfg.profile_phases and state.device.type == "cuda":
+            torch.cuda.synchronize()
+        return time.perf_counter()
+
     offsets, signs = state.member_offsets_signs(state.generation)
     pop_w = state.perturb_local(offsets, signs)                      # K2
+    t_perturb = _phase_mark()
     fitness, env_steps, ep_len = state.rollout(pop_w)                # K7/K3/K5/K8
+    t_rollout = _phase_mark()
 
     # C3 rides the side HIP stream (overlap_comm) while the host derives the
     # NEXT generation's member offsets (BASELINE: collectives overlapped with
@@ -321,6 +329,10 @@ def train_step(state: TrainState) -> dict:
         "steps_per_sec": env_steps_global / dt,
         "gen_time_s": dt,
     }
+    if cfg.profile_phases:
+        stats["t_perturb_ms"] = (t_perturb - t0) * 1e3
+        stats["t_rollout_ms"] = (t_rollout - t_perturb) * 1e3
+        stats["t_update_ms"] = dt * 1e3 - stats["t_perturb_ms"] - stats["t_rollout_ms"]
     if cfg.checkpoint_path and cfg.checkpoint_every and state.rank == 0 \
             and state.generation % cfg.checkpoint_every == 0:
         save_checkpoint(cfg.checkpoint_path, state.checkpoint_payload())

@@ -141,3 +141,9 @@ def test_sgd_optimizer_runs():
     train_step(st)
     assert st.optimizer.t == 1
     assert torch.isfinite(st.theta).all()
+
+
+def test_phase_timings_in_stats():
+    st = TrainState(_mini("es", profile_phases=True), ops_backend="torch_ref")
+    s = train_step(st)
+    assert s["t_perturb_ms"] >= 0 and s["t_rollout_ms"] > 0
