@@ -112,10 +112,13 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
     __syncthreads();
   }
 
-  // epilogue: bias (+skip) -> global, D: lane l reg r -> (m, n)
+  // epilogue: the naive per-lane scatter hits a 2*HW*HW-byte stride (one
+  // cache line per lane) — stage the bias+relu'd tile in LDS (sa is free
+  // and exactly CO*M elements) and do a coalesced cooperative store instead
   const uint16_t* bias = w + b_off;
   uint16_t* dst = out + (int64_t)i * CO * HW * HW;
   const uint16_t* sk = ADD_SKIP ? skip + (int64_t)i * CO * HW * HW : nullptr;
+  static_assert(CO * M <= M * 32, "output tile exceeds the sa staging carve");
 #pragma unroll
   for (int u = 0; u < MAXT; ++u) {
     int t = wid + NWAVES * u;
@@ -124,13 +127,16 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
     int m0 = mt * 16 + ((lane >> 4) << 2);
     float b = des::bf16_to_f32(bias[n]);
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      int m = m0 + r;
-      int oy = ty0 + m / TS, ox = tx0 + m % TS;
-      float v = acc[u][r] + b;
-      if (ADD_SKIP) v += des::bf16_to_f32(sk[(n * HW + oy) * HW + ox]);
-      dst[(n * HW + oy) * HW + ox] = des::f32_to_bf16(v);
-    }
+    for (int r = 0; r < 4; ++r)
+      sa[n * M + m0 + r] = des::f32_to_bf16(acc[u][r] + b);
+  }
+  __syncthreads();
+  for (int e = tid; e < CO * M; e += M) {
+    int n = e / M, m = e - n * M;
+    int64_t gidx = ((int64_t)n * HW + ty0 + m / TS) * HW + tx0 + m % TS;
+    float v = des::bf16_to_f32(sa[e]);
+    if (ADD_SKIP) v += des::bf16_to_f32(sk[gidx]);
+    dst[gidx] = des::f32_to_bf16(v);
   }
 }
 
