@@ -1,0 +1,48 @@
+"""Config/engine validation and error-path tests."""
+import pytest
+import torch
+
+from distributedes_amd import preset
+from distributedes_amd.config import Config
+from distributedes_amd.core.engine import TrainState
+
+
+def test_antithetic_odd_pop_rejected():
+    with pytest.raises(ValueError, match="even population"):
+        Config(pop=7, antithetic=True).validate()
+
+
+def test_unknown_algo_rejected():
+    with pytest.raises(ValueError, match="algo"):
+        Config(algo="genetic").validate()
+
+
+def test_unknown_env_model_rejected():
+    from distributedes_amd.envs import make_env
+    with pytest.raises(KeyError):
+        make_env("mujoco_real")
+    with pytest.raises(KeyError):
+        preset("nonexistent_preset")
+
+
+def test_pop_not_divisible_by_world(monkeypatch):
+    import distributedes_amd.parallel.comm as comm_mod
+
+    class FakeComm:
+        world, rank, backend = 3, 0, None
+        is_dist = True
+        device = torch.device("cpu")
+        def broadcast_theta(self, t): pass
+        def broadcast_seed(self, s): return s
+    cfg = preset("cartpole_es", pop=8, noise_table_size=1 << 16, device="cpu")
+    with pytest.raises(ValueError, match="divisible"):
+        TrainState(cfg, comm=FakeComm(), ops_backend="torch_ref")
+
+
+def test_table_smaller_than_dim_rejected():
+    cfg = preset("cartpole_es", pop=4, noise_table_size=256, device="cpu")
+    st = None
+    with pytest.raises(AssertionError, match="noise table"):
+        st = TrainState(cfg, ops_backend="torch_ref")
+        from distributedes_amd.core.engine import train_step
+        train_step(st)
