@@ -155,8 +155,12 @@ class TrainState:
         P = pop_w.shape[0]
         gen_seed = noise_mod.splitmix64(cfg.seed ^ (self.generation * 0x9E37)) & 0x7FFFFFFF
         if self._hip_rollout is not None:
+            # hipGraph capture only for the persistent training buffer:
+            # transient pop_w (evaluate()) must not bake freed pointers into
+            # a cached graph
             return self._hip_rollout.run(pop_w, gen_seed, self.env_ids(),
-                                         cfg.episode_steps, self.normalizer)
+                                         cfg.episode_steps, self.normalizer,
+                                         allow_graph=pop_w is self._pop_w)
         if pop_w.shape[1] != self.dim:           # padded rows -> logical view
             pop_w = pop_w[:, :self.dim]
         obs = self.env.reset(P, gen_seed, device=self.device, env_ids=self.env_ids())

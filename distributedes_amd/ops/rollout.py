@@ -275,13 +275,14 @@ class HipRollout:
             self._env_step(slot)
 
     def run(self, pop_w: torch.Tensor, gen_seed: int, env_ids: torch.Tensor,
-            episode_steps: int, normalizer):
+            episode_steps: int, normalizer, allow_graph: bool = True):
         P = pop_w.shape[0]
         if self.P != P:
             self._alloc(P)
         is_state_env = len(self.env.obs_shape) == 1
         self._reset(gen_seed, env_ids)
-        use_graph = (getattr(self.cfg, "use_hip_graphs", False)
+        use_graph = (allow_graph
+                     and getattr(self.cfg, "use_hip_graphs", False)
                      and self.device.type == "cuda"
                      and not getattr(self, "_graphs_broken", False))
         if use_graph:
