@@ -23,6 +23,8 @@
 // With 2 blocks/CU one block's (now tiny) conv phase and the other block's
 // HBM-bound FC stream overlap; the step floor is the FC traffic:
 // pop-8192 x 3.2 MiB / 6.3 TB/s ~= 4.3 ms.
+#include <cstdlib>
+
 #include "common.h"
 #include "gemv.h"
 
@@ -102,7 +104,7 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
     const IT* __restrict__ frames,     // [P][4][84][84]
     const int* __restrict__ slot_map,  // [4] logical->physical slot
     int* __restrict__ actions, float* __restrict__ logits_out, int A, int P,
-    const int64_t* __restrict__ offs) {
+    const int64_t* __restrict__ offs, unsigned stagger) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint16_t* sa = reinterpret_cast<uint16_t*>(smem + OFF_SA);
   uint16_t* sb = reinterpret_cast<uint16_t*>(smem + OFF_SB);
@@ -114,6 +116,17 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
 
   const int i = blockIdx.x;
   if (i >= P) return;
+  // phase stagger: the two co-resident blocks of a CU otherwise run their
+  // conv (compute) and FC (HBM) phases in lockstep, so conv time never
+  // hides under the FC stream.  Delaying the odd block of the FIRST
+  // resident round by ~one conv phase anti-phases the whole kernel; block
+  // retirement then sustains the offset (speed heuristic only — dispatch
+  // order is not a correctness assumption here).
+  if (stagger && (blockIdx.x & 1) && blockIdx.x < 1024) {
+    unsigned long long t0 = __builtin_amdgcn_s_memtime();
+    while (__builtin_amdgcn_s_memtime() - t0 < (unsigned long long)stagger)
+      __builtin_amdgcn_s_sleep(16);
+  }
   const uint16_t* w = popw + (int64_t)i * row_stride;
   const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
   const IT* obs = frames + (int64_t)i * (4 * HW * HW);
@@ -296,6 +309,10 @@ extern "C" void des_launch_nature_cnn_forward(
     const uint16_t* popw, int64_t row_stride, const void* frames, bool obs_u8,
     const int* slot_map_dev, int* actions, float* logits_out, int A, int P,
     const int64_t* offs_dev, hipStream_t stream) {
+  static unsigned stagger = [] {
+    const char* e = getenv("DES_NATURE_STAGGER");
+    return e ? (unsigned)atoi(e) : 0u;
+  }();
   static bool attr_set = false;
   if (!attr_set) {
     hipFuncSetAttribute((const void*)nature_conv_kernel<uint8_t>,
@@ -308,10 +325,10 @@ extern "C" void des_launch_nature_cnn_forward(
     hipLaunchKernelGGL(nature_conv_kernel<uint8_t>, dim3((uint32_t)P),
                        dim3(512), CONV_LDS, stream, popw, row_stride,
                        (const uint8_t*)frames, slot_map_dev, actions,
-                       logits_out, A, P, offs_dev);
+                       logits_out, A, P, offs_dev, stagger);
   else
     hipLaunchKernelGGL(nature_conv_kernel<uint16_t>, dim3((uint32_t)P),
                        dim3(512), CONV_LDS, stream, popw, row_stride,
                        (const uint16_t*)frames, slot_map_dev, actions,
-                       logits_out, A, P, offs_dev);
+                       logits_out, A, P, offs_dev, stagger);
 }
