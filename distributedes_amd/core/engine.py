@@ -203,10 +203,21 @@ class TrainState:
         return self._pop_w
 
     # ------------------------------------------------------------------ #
+    def shape_fitness(self, global_fitness: torch.Tensor) -> torch.Tensor:
+        """K9: centered-rank kernel on the HIP path (bit-identical integer
+        counting), torch reference elsewhere/for other modes."""
+        cfg = self.cfg
+        if (cfg.rank_shaping == "centered" and self.ops.name == "hip"
+                and global_fitness.is_cuda and global_fitness.numel() >= 2):
+            u = torch.empty_like(global_fitness)
+            self.ops.ext.centered_rank(global_fitness.contiguous(), u)
+            return u
+        return shape_fitness(global_fitness, cfg.rank_shaping)
+
     def es_update(self, global_fitness: torch.Tensor, offsets, signs):
         """Centered-rank weighted noise sum (K10) + C4 all-reduce + Adam."""
         cfg = self.cfg
-        u = shape_fitness(global_fitness, cfg.rank_shaping)          # K9, all ranks
+        u = self.shape_fitness(global_fitness)                       # K9, all ranks
         lo, hi = self.member_base, self.member_base + self.local_pop
         w_local = u[lo:hi].to(self.device)
         partial = self.ops.weighted_noise_sum(self.table, offsets, signs,

@@ -49,6 +49,7 @@ void des_launch_humanoid_step(float*, int*, const float*, float*, uint8_t*,
                               float*, int, int, hipStream_t);
 void des_launch_normalizer_observe(const float*, double*, double*, int, int,
                                    hipStream_t);
+void des_launch_centered_rank(const float*, float*, int, hipStream_t);
 void des_launch_impala_conv3(const uint16_t*, uint16_t*, const uint16_t*,
                              const uint16_t*, int64_t, int64_t, int64_t, int,
                              bool, bool, int, hipStream_t);
@@ -312,6 +313,14 @@ void impala_fc_head(torch::Tensor popw, torch::Tensor act,
                             cur_stream());
 }
 
+void centered_rank(torch::Tensor fitness, torch::Tensor u) {
+  CHECK_DEV(fitness); CHECK_CONT(fitness); CHECK_DEV(u); CHECK_CONT(u);
+  TORCH_CHECK(fitness.dtype() == torch::kFloat32 && u.numel() == fitness.numel());
+  TORCH_CHECK(fitness.numel() >= 2);
+  des_launch_centered_rank(cfptr(fitness), fptr(u), (int)fitness.numel(),
+                           cur_stream());
+}
+
 void normalizer_observe(torch::Tensor obs, torch::Tensor sum,
                         torch::Tensor sumsq) {
   CHECK_DEV(obs); CHECK_CONT(obs);
@@ -341,6 +350,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("humanoid_reset", &humanoid_reset);
   m.def("humanoid_step", &humanoid_step);
   m.def("normalizer_observe", &normalizer_observe);
+  m.def("centered_rank", &centered_rank, "K9: stable centered-rank weights");
   m.def("impala_conv", &impala_conv, "K6: 3x3 pad-1 conv (pop, tiled)");
   m.def("impala_pool", &impala_pool, "K6: 3x3 s2 pad-1 max pool");
   m.def("impala_fc_head", &impala_fc_head, "K6: fc 2048->256 + head + argmax");

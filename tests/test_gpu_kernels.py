@@ -119,3 +119,15 @@ def test_bf16_table_fill_and_perturb(ops, ref, table_dtype):
     acc = ops.weighted_noise_sum(gpu, offsets, signs,
                                  torch.rand(8, device="cuda"), D)
     assert torch.isfinite(acc).all()
+
+
+def test_centered_rank_kernel_bitexact(ops):
+    from distributedes_amd.core.rank import centered_ranks
+    torch.manual_seed(0)
+    for n in (2, 17, 1000, 8192):
+        f = torch.randn(n, device="cuda")
+        f[:: max(1, n // 7)] = 1.25   # force ties
+        u = torch.empty_like(f)
+        ops.ext.centered_rank(f, u)
+        ref = centered_ranks(f.cpu())
+        assert torch.equal(u.cpu(), ref)   # integer counting: bit-exact
