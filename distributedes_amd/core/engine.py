@@ -278,8 +278,9 @@ class TrainState:
         rng = payload.get("rng", {})
         if rng.get("table_seed") is not None and rng["table_seed"] != self.table_seed:
             self.table_seed = int(rng["table_seed"])
-            self.table = self.ops.fill_table(self.cfg.noise_table_size,
-                                             self.table_seed, self.device)
+            self.table = self.ops.fill_table(
+                self.cfg.noise_table_size, self.table_seed, self.device,
+                dtype=self.table.dtype)
         if self.cfg.algo == "cem" and "algo" in payload:
             self.sigma_vec = payload["algo"]["sigma_vec"].to(self.device)
             self.cem_noise_floor = float(payload["algo"]["noise_floor"])

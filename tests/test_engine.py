@@ -147,3 +147,18 @@ def test_phase_timings_in_stats():
     st = TrainState(_mini("es", profile_phases=True), ops_backend="torch_ref")
     s = train_step(st)
     assert s["t_perturb_ms"] >= 0 and s["t_rollout_ms"] > 0
+
+
+def test_resume_foreign_seed_keeps_table_dtype():
+    """Loading a checkpoint whose table_seed differs must rebuild the table
+    in the configured noise dtype."""
+    cfg = _mini("es").replace(noise_dtype="bf16")
+    a = TrainState(cfg, ops_backend="torch_ref")
+    train_step(a)
+    payload = a.checkpoint_payload()
+    payload["rng"]["table_seed"] = 12345  # pretend another run's table
+    b = TrainState(cfg, ops_backend="torch_ref")
+    b.load_payload(payload)
+    assert b.table.dtype == torch.bfloat16
+    assert b.table_seed == 12345
+    train_step(b)
