@@ -42,6 +42,8 @@ def main():
     ap.add_argument("--profile", default=None, metavar="DIR",
                     help="write a torch.profiler chrome trace of the timed "
                          "steps into DIR (SURVEY.md §5.1 tracing hook)")
+    ap.add_argument("--set", nargs="*", default=[], metavar="K=V",
+                    help="extra Config overrides, e.g. --set noise_dtype='bf16'")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -62,6 +64,13 @@ def main():
         overrides["noise_table_size"] = args.table
     if args.backend:
         overrides["backend"] = args.backend
+    import ast
+    for kv in args.set:
+        k, v = kv.split("=", 1)
+        try:
+            overrides[k] = ast.literal_eval(v)
+        except (ValueError, SyntaxError):
+            overrides[k] = v
     if not on_gpu:
         # CPU smoke only (the driver benches on a GPU box): shrink hard
         overrides.setdefault("pop", 8)
