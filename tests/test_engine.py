@@ -162,3 +162,18 @@ def test_resume_foreign_seed_keeps_table_dtype():
     assert b.table.dtype == torch.bfloat16
     assert b.table_seed == 12345
     train_step(b)
+
+
+def test_render_filmstrip(tmp_path):
+    """SURVEY §3.3 eval/render stack: rollout frames -> SVG filmstrip."""
+    from distributedes_amd import run
+    from distributedes_amd.utils.render import frames_to_svg, rollout_frames
+    p = str(tmp_path / "ck.pt")
+    cfg = preset("pong_es_8k", pop=4, episode_steps=6, generations=1,
+                 noise_table_size=1 << 22, device="cpu",
+                 checkpoint_path=p, log_every=10 ** 9)
+    run(cfg, ops_backend="torch_ref")
+    frames, rewards, actions = rollout_frames(cfg, p, n_frames=3, stride=2)
+    assert len(frames) == 3 and frames[0].shape == (84, 84)
+    svg = frames_to_svg(frames, rewards)
+    assert "<svg" in svg
