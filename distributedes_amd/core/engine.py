@@ -250,6 +250,14 @@ class TrainState:
         self.cem_noise_floor *= cfg.cem_noise_decay
         self.sigma_vec = (self.sigma_vec ** 2 * var_eps
                           + self.cem_noise_floor ** 2).sqrt()
+        if (cfg.theta_allreduce_every
+                and (self.generation + 1) % cfg.theta_allreduce_every == 0):
+            # drift insurance for mu AND the diagonal sigma (same role as the
+            # ES path's periodic parameter all-reduce)
+            buf2 = torch.cat([self.theta, self.sigma_vec])
+            self.comm.allreduce_(buf2, avg=True)
+            self.theta = buf2[:self.dim].clone()
+            self.sigma_vec = buf2[self.dim:].clone()
 
     # ------------------------------------------------------------------ #
     def checkpoint_payload(self) -> dict:
