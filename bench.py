@@ -141,6 +141,8 @@ def main():
                 "sigma": cfg.sigma,
                 "generations_per_sec": gens_per_sec,
                 "ops_backend": state.ops.name,
+                "hip_graphs": bool(getattr(state._hip_rollout,
+                                           "last_used_graph", False)),
             },
         }
         print(json.dumps(rec), flush=True)

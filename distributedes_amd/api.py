@@ -77,17 +77,13 @@ def evaluate(checkpoint_or_state, config: Optional[Config] = None,
     stride = ((state.dim + 7) & ~7) if state._hip_rollout is not None else state.dim
     pop_w = torch.zeros(P, stride, dtype=dt, device=state.device)
     pop_w[:, :state.dim] = state.theta.to(dt)
-    # distinct env streams per episode (antithetic pairing would make
-    # episodes 2k and 2k+1 share a stream)
-    saved_base, saved_pop = state.member_base, state.local_pop
-    saved_cfg = state.cfg
-    state.member_base, state.local_pop = 0, P
-    state.cfg = state.cfg.replace(antithetic=False)
-    try:
-        fitness, env_steps, ep_len = state.rollout(pop_w)
-    finally:
-        state.member_base, state.local_pop = saved_base, saved_pop
-        state.cfg = saved_cfg
+    # scoped eval: explicit env ids (one distinct stream per episode — the
+    # training slice would pair-share streams under antithetic sampling) and
+    # observe=False so eval observations never enter the normalizer deltas.
+    # No TrainState fields are touched.
+    env_ids = torch.arange(P, dtype=torch.int64)
+    fitness, env_steps, ep_len = state.rollout(pop_w, env_ids=env_ids,
+                                               observe=False)
     return {
         "return_mean": float(fitness.mean()),
         "return_max": float(fitness.max()),

@@ -20,7 +20,9 @@ accumulated nondeterminism.
 """
 from __future__ import annotations
 
+import logging
 import math
+import os
 import time
 from typing import Optional
 
@@ -36,11 +38,22 @@ from . import noise as noise_mod
 from .optim import make_optimizer
 from .rank import shape_fitness
 
+log = logging.getLogger("distributedes_amd")
+
 
 def _resolve_device(cfg: Config) -> torch.device:
     if cfg.device == "auto":
-        return torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
-    return torch.device(cfg.device)
+        dev = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+    else:
+        dev = torch.device(cfg.device)
+    # under torchrun each rank must land on its own GPU BEFORE the process
+    # group initializes (else every rank grabs GPU 0 and NCCL hangs/errors)
+    if dev.type == "cuda" and dev.index is None:
+        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+        if local_rank < torch.cuda.device_count():
+            dev = torch.device("cuda", local_rank)
+            torch.cuda.set_device(dev)
+    return dev
 
 
 def _make_ops(device: torch.device, want_hip: bool = True):
@@ -117,6 +130,11 @@ class TrainState:
             if HipRollout.supports(cfg, self.model):
                 self._hip_rollout = HipRollout(cfg, self.env, self.model,
                                                self.ops.ext, self.device)
+            else:
+                log.warning(
+                    "env=%s model=%s precision=%s has no HIP rollout path; "
+                    "the hot loop will run eager torch ops on GPU",
+                    cfg.env, type(self.model).__name__, cfg.precision)
 
     # ------------------------------------------------------------------ #
     def member_offsets_signs(self, generation: int):
@@ -148,22 +166,31 @@ class TrainState:
         return gids
 
     # ------------------------------------------------------------------ #
-    def rollout(self, pop_w: torch.Tensor):
+    def rollout(self, pop_w: torch.Tensor, env_ids: Optional[torch.Tensor] = None,
+                observe: bool = True):
         """Evaluate local members; returns (fitness [P] fp32, env_steps int,
-        ep_len_mean float)."""
+        ep_len_mean float).
+
+        env_ids: env-stream ids (default: the training member slice).
+        observe=False keeps the rollout's observations OUT of the running
+        normalizer deltas — evaluation must not perturb training statistics.
+        """
         cfg = self.cfg
         P = pop_w.shape[0]
+        if env_ids is None:
+            env_ids = self.env_ids()
         gen_seed = noise_mod.splitmix64(cfg.seed ^ (self.generation * 0x9E37)) & 0x7FFFFFFF
         if self._hip_rollout is not None:
             # hipGraph capture only for the persistent training buffer:
             # transient pop_w (evaluate()) must not bake freed pointers into
             # a cached graph
-            return self._hip_rollout.run(pop_w, gen_seed, self.env_ids(),
+            return self._hip_rollout.run(pop_w, gen_seed, env_ids,
                                          cfg.episode_steps, self.normalizer,
-                                         allow_graph=pop_w is self._pop_w)
+                                         allow_graph=pop_w is self._pop_w,
+                                         observe=observe)
         if pop_w.shape[1] != self.dim:           # padded rows -> logical view
             pop_w = pop_w[:, :self.dim]
-        obs = self.env.reset(P, gen_seed, device=self.device, env_ids=self.env_ids())
+        obs = self.env.reset(P, gen_seed, device=self.device, env_ids=env_ids)
         fitness = torch.zeros(P, dtype=torch.float32, device=self.device)
         alive = torch.ones(P, dtype=torch.bool, device=self.device)
         ep_len = torch.zeros(P, dtype=torch.float32, device=self.device)
@@ -171,7 +198,8 @@ class TrainState:
         steps_done = 0
         for t in range(cfg.episode_steps):
             if is_state_env:
-                self.normalizer.observe(obs)
+                if observe:
+                    self.normalizer.observe(obs)
                 x = self.normalizer.normalize(obs)
             else:
                 x = obs
@@ -274,7 +302,7 @@ class TrainState:
             "rng": {"run_seed": self.cfg.seed, "table_seed": self.table_seed},
         }
         if self.cfg.algo == "cem":
-            payload["algo"] = {"sigma_vec": self.sigma_vec.cpu(),
+            payload["algo"] = {"sigma_vec": self.sigma_vec.cpu().clone(),
                                "noise_floor": self.cem_noise_floor}
         return payload
 
@@ -352,6 +380,8 @@ def train_step(state: TrainState) -> dict:
         "env_steps_total": state.env_steps_total,
         "steps_per_sec": env_steps_global / dt,
         "gen_time_s": dt,
+        "ops_backend": state.ops.name,
+        "hip_graphs": bool(getattr(state._hip_rollout, "last_used_graph", False)),
     }
     if cfg.profile_phases:
         stats["t_perturb_ms"] = (t_perturb - t0) * 1e3

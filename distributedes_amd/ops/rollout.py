@@ -10,7 +10,11 @@ host<->device transfers and no torch ops, so it is hipGraph-capturable
 """
 from __future__ import annotations
 
+import logging
+
 import torch
+
+log = logging.getLogger("distributedes_amd")
 
 from ..models import ImpalaCNNPolicy, MLPPolicy, NatureCNNPolicy
 from ..models.base import ParamSpec
@@ -101,6 +105,11 @@ class HipRollout:
         z = lambda *s, dt=torch.float32: torch.zeros(*s, dtype=dt, device=dev)  # noqa: E731
         self._empty = torch.empty(0, device=dev)
         self.seed_buf = torch.zeros(1, dtype=torch.int64, device=dev)
+        # persistent env-id buffer: captured hipGraphs bake this pointer into
+        # the step kernels, so it must never be reallocated between replays
+        # (a fresh .to(device) tensor each generation gets freed and its block
+        # reused, silently corrupting per-env RNG streams under graph replay)
+        self._ids = torch.zeros(P, dtype=torch.int64, device=dev)
         self._graphs = {}
         self.fitness = z(P)
         self.alive = z(P, dt=torch.uint8)
@@ -149,7 +158,8 @@ class HipRollout:
     # ------------------------------------------------------------------ #
     def _reset(self, gen_seed: int, env_ids: torch.Tensor):
         e = self.env_name
-        ids = env_ids.to(self.device)
+        self._ids.copy_(env_ids.to(self.device))
+        ids = self._ids
         if e == "cartpole":
             self.ext.cartpole_reset(self.st_state, self.st_steps, self.st_rctr,
                                     self.fitness, self.alive, self.ep_len,
@@ -167,7 +177,6 @@ class HipRollout:
         elif e == "humanoid_synth":
             self.ext.humanoid_reset(self.hu_s, self.hu_steps, self.fitness,
                                     self.alive, self.ep_len, ids, gen_seed)
-        self._ids = ids
         self._gen_seed = gen_seed
         # the step kernels read their hash seed from device memory so a
         # captured hipGraph replays correctly across generations
@@ -264,18 +273,20 @@ class HipRollout:
         self.ext.normalizer_observe(obs, normalizer._dsum, normalizer._dsumsq)
 
     # ------------------------------------------------------------------ #
-    def _loop(self, pop_w, episode_steps, normalizer, is_state_env):
+    def _loop(self, pop_w, episode_steps, normalizer, is_state_env,
+              observe=True):
         # pong ring: reset wrote the initial frame to every slot; step t
         # writes slot t % 4 and the forward sees slots in logical order
         for t in range(episode_steps):
             slot = t % 4
-            if is_state_env:
+            if is_state_env and observe:
                 self._observe(normalizer)
             self._forward(pop_w, normalizer, slot_newest=(t - 1) % 4)
             self._env_step(slot)
 
     def run(self, pop_w: torch.Tensor, gen_seed: int, env_ids: torch.Tensor,
-            episode_steps: int, normalizer, allow_graph: bool = True):
+            episode_steps: int, normalizer, allow_graph: bool = True,
+            observe: bool = True):
         P = pop_w.shape[0]
         if self.P != P:
             self._alloc(P)
@@ -286,26 +297,31 @@ class HipRollout:
                      and self.device.type == "cuda"
                      and not getattr(self, "_graphs_broken", False))
         if use_graph:
-            key = (episode_steps, pop_w.data_ptr(), id(normalizer))
+            key = (episode_steps, pop_w.data_ptr(), id(normalizer), observe)
             g = self._graphs.get(key)
             if g is None:
                 try:
                     g = self._capture(pop_w, gen_seed, env_ids, episode_steps,
-                                      normalizer, is_state_env)
+                                      normalizer, is_state_env, observe)
                     self._graphs[key] = g
                 except Exception:
+                    log.warning(
+                        "hipGraph capture failed; falling back to eager "
+                        "kernel launches for all subsequent generations",
+                        exc_info=True)
                     self._graphs_broken = True
                     use_graph = False
             if use_graph:
                 g.replay()
         if not use_graph:
-            self._loop(pop_w, episode_steps, normalizer, is_state_env)
-        if is_state_env and hasattr(normalizer, "_dcount"):
+            self._loop(pop_w, episode_steps, normalizer, is_state_env, observe)
+        self.last_used_graph = use_graph
+        if is_state_env and observe and hasattr(normalizer, "_dcount"):
             normalizer._dcount += P * episode_steps
         return self.fitness.clone(), P * episode_steps, float(self.ep_len.mean())
 
     def _capture(self, pop_w, gen_seed, env_ids, episode_steps, normalizer,
-                 is_state_env):
+                 is_state_env, observe=True):
         """Capture one generation's kernel sequence into a hipGraph.
         The warmup pass EXECUTES a full loop (torch requirement), so env +
         normalizer-delta state is re-reset before capture; the replay then
@@ -313,7 +329,7 @@ class HipRollout:
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
-            self._loop(pop_w, episode_steps, normalizer, is_state_env)
+            self._loop(pop_w, episode_steps, normalizer, is_state_env, observe)
         torch.cuda.current_stream().wait_stream(s)
         # undo warmup side effects
         self._reset(self._gen_seed, self._ids)
@@ -322,5 +338,5 @@ class HipRollout:
             normalizer._dsumsq.zero_()
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
-            self._loop(pop_w, episode_steps, normalizer, is_state_env)
+            self._loop(pop_w, episode_steps, normalizer, is_state_env, observe)
         return g

@@ -23,6 +23,17 @@ sources = sorted(str(p) for p in CSRC.glob("*.cpp")) + \
           sorted(str(p) for p in CSRC.glob("*.hip")
                  if not p.name.endswith("_hip.hip"))  # exclude hipify copies
 
+class CleanHipifyBuildExt(BuildExtension):
+    """Delete torch-hipify's `*_hip.hip` byproducts after building: they are
+    verbatim copies of the sources (which are already native HIP) and only
+    pollute the tree / gpurun snapshots."""
+
+    def run(self):
+        super().run()
+        for p in CSRC.glob("*_hip.hip"):
+            p.unlink()
+
+
 setup(
     name="distributedes_amd_hip",
     ext_modules=[
@@ -35,5 +46,5 @@ setup(
             },
         )
     ],
-    cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
+    cmdclass={"build_ext": CleanHipifyBuildExt.with_options(no_python_abi_suffix=False)},
 )
