@@ -383,6 +383,18 @@ def train_step(state: TrainState) -> dict:
         "ops_backend": state.ops.name,
         "hip_graphs": bool(getattr(state._hip_rollout, "last_used_graph", False)),
     }
+    ng = getattr(state.env, "n_games", 1)
+    if ng > 1:
+        # per-game fitness breakdown (BASELINE config #4: Atari-57 means 57
+        # tasks in one population; game = env_id % n_games)
+        m = torch.arange(cfg.pop, device=global_fitness.device)
+        game = ((m // 2 if cfg.antithetic else m) % ng)
+        sums = torch.zeros(ng, device=global_fitness.device).index_add_(
+            0, game, global_fitness)
+        cnts = torch.zeros(ng, device=global_fitness.device).index_add_(
+            0, game, torch.ones_like(global_fitness))
+        per_game = sums / cnts.clamp_min(1.0)
+        stats["per_game_return_mean"] = [round(float(x), 4) for x in per_game]
     if cfg.profile_phases:
         stats["t_perturb_ms"] = (t_perturb - t0) * 1e3
         stats["t_rollout_ms"] = (t_rollout - t_perturb) * 1e3

@@ -39,10 +39,10 @@ void des_launch_pong_step(float*, float*, float*, float*, float*, float*,
                           const int64_t*, int, hipStream_t);
 void des_launch_latent_reset(float*, int*, float*, uint8_t*, float*,
                              const int64_t*, int64_t, int, hipStream_t);
-void des_launch_latent_render(const float*, uint16_t*, int, int, int,
-                              hipStream_t);
-void des_launch_latent_step(float*, int*, const int*, float*, uint8_t*,
-                            float*, int, int, int, hipStream_t);
+void des_launch_latent_render(const float*, const float*, uint16_t*, int, int,
+                              int, hipStream_t);
+void des_launch_latent_step(float*, const float*, int*, const int*, float*,
+                            uint8_t*, float*, int, int, int, hipStream_t);
 void des_launch_humanoid_reset(float*, int*, float*, uint8_t*, float*,
                                const int64_t*, int64_t, int, hipStream_t);
 void des_launch_humanoid_step(float*, int*, const float*, float*, uint8_t*,
@@ -244,20 +244,24 @@ void latent_reset(torch::Tensor z, torch::Tensor steps, torch::Tensor fitness,
                           cur_stream());
 }
 
-void latent_render(torch::Tensor z, torch::Tensor obs) {
+void latent_render(torch::Tensor z, torch::Tensor coef, torch::Tensor obs) {
   TORCH_CHECK(obs.dtype() == torch::kBFloat16);
+  TORCH_CHECK(coef.size(0) == z.size(0) && coef.size(1) == 6);
   int C = obs.size(1), HW = obs.size(2);
-  des_launch_latent_render(cfptr(z),
+  des_launch_latent_render(cfptr(z), cfptr(coef),
                            reinterpret_cast<uint16_t*>(obs.data_ptr()), C, HW,
                            z.size(0), cur_stream());
 }
 
-void latent_step(torch::Tensor z, torch::Tensor steps, torch::Tensor actions,
-                 torch::Tensor fitness, torch::Tensor alive,
-                 torch::Tensor ep_len, int64_t num_actions, int64_t horizon) {
-  des_launch_latent_step(fptr(z), iptr(steps), iptr(actions), fptr(fitness),
-                         bptr(alive), fptr(ep_len), (int)num_actions,
-                         (int)horizon, z.size(0), cur_stream());
+void latent_step(torch::Tensor z, torch::Tensor coef, torch::Tensor steps,
+                 torch::Tensor actions, torch::Tensor fitness,
+                 torch::Tensor alive, torch::Tensor ep_len,
+                 int64_t num_actions, int64_t horizon) {
+  TORCH_CHECK(coef.size(0) == z.size(0) && coef.size(1) == 6);
+  des_launch_latent_step(fptr(z), cfptr(coef), iptr(steps), iptr(actions),
+                         fptr(fitness), bptr(alive), fptr(ep_len),
+                         (int)num_actions, (int)horizon, z.size(0),
+                         cur_stream());
 }
 
 void humanoid_reset(torch::Tensor s, torch::Tensor steps,

@@ -29,8 +29,12 @@ __global__ void latent_reset_kernel(float* __restrict__ z,
   steps[i] = 0; fitness[i] = 0.f; alive[i] = 1; ep_len[i] = 0.f;
 }
 
+// coef: [P, 6] per-env game coefficients (A, B, C, RF, RW, F0) computed on
+// the host by envs/synthetic.py game_coefs (single source of truth — the
+// Atari-57 multi-task parameterization; single-game envs get the classics).
 template <int C, int HW>
 __global__ void latent_render_kernel(const float* __restrict__ z,
+                                     const float* __restrict__ coef,
                                      uint16_t* __restrict__ obs, int P) {
   int i = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   int lane = threadIdx.x & 63;
@@ -38,12 +42,13 @@ __global__ void latent_render_kernel(const float* __restrict__ z,
   float zl[ZDIM];
 #pragma unroll
   for (int j = 0; j < ZDIM; ++j) zl[j] = z[i * ZDIM + j];
+  float f0 = coef[i * 6 + 5];
   uint16_t* out = obs + (int64_t)i * C * HW * HW;
   for (int pix = lane; pix < C * HW * HW; pix += 64) {
     int c = pix / (HW * HW), r = pix % (HW * HW);
     int h = r / HW, w = r % HW;
     float zc0 = zl[c % ZDIM], zc1 = zl[(c + 1) % ZDIM], zc2 = zl[(c + 2) % ZDIM];
-    float v = 0.5f + 0.5f * __sinf(3.0f * zc0 + 0.07f * h * (1.0f + zc1) +
+    float v = 0.5f + 0.5f * __sinf(f0 * zc0 + 0.07f * h * (1.0f + zc1) +
                                    0.05f * w * (1.0f + zc2));
     out[pix] = des::f32_to_bf16(v);
   }
@@ -51,6 +56,7 @@ __global__ void latent_render_kernel(const float* __restrict__ z,
 
 template <int NA>
 __global__ void latent_step_kernel(float* __restrict__ z,
+                                   const float* __restrict__ coef,
                                    int* __restrict__ steps,
                                    const int* __restrict__ actions,
                                    float* __restrict__ fitness,
@@ -61,18 +67,20 @@ __global__ void latent_step_kernel(float* __restrict__ z,
   if (i >= P) return;
   float a = (float)actions[i];
   float phase = a * (6.2831853071795864f / NA);
+  float cA = coef[i * 6 + 0], cB = coef[i * 6 + 1], cC = coef[i * 6 + 2];
+  float cRF = coef[i * 6 + 3], cRW = coef[i * 6 + 4];
   float zo[ZDIM], zn[ZDIM];
 #pragma unroll
   for (int j = 0; j < ZDIM; ++j) zo[j] = z[i * ZDIM + j];
 #pragma unroll
   for (int j = 0; j < ZDIM; ++j) {
     float zr = zo[(j + ZDIM - 1) % ZDIM];  // torch.roll(z, 1)
-    zn[j] = tanhf(1.3f * zo[j] + 0.7f * zr * cosf(phase) +
-                  0.15f * sinf(phase + zr));
+    zn[j] = tanhf(cA * zo[j] + cB * zr * cosf(phase) +
+                  cC * sinf(phase + zr));
     z[i * ZDIM + j] = zn[j];
   }
   int st = steps[i] + 1;
-  float reward = sinf(3.0f * zn[0] + 0.5f * a);
+  float reward = sinf(cRF * zn[0] + cRW * a);
   if (alive[i]) { fitness[i] += reward; ep_len[i] += 1.0f; }
   if (st >= horizon) st = 0;  // horizon handled by rollout; keep counter sane
   steps[i] = st;
@@ -116,19 +124,27 @@ __global__ void humanoid_step_kernel(float* __restrict__ s,
   if (lane < HACT) areg[lane] = arow[lane];
   __syncthreads();
 
+  // target-tracking reward (== envs/synthetic.py HumanoidSyntheticEnv.step):
+  // the first 17 dims chase a time-varying reference trajectory, so ES
+  // learning curves rise over hundreds of generations (no trivial ceiling)
+  float t = (float)(steps[i] + 1);
   float rpart = 0.f, apart = 0.f;
   for (int j = lane; j < HDIM; j += 64) {
     float sr = sold[(j + HDIM - 1) % HDIM];
     float u = areg[j % HACT];
     float sn = tanhf(1.2f * sold[j] + 0.6f * sr + 0.3f * u);
     srow[j] = sn;
-    if (j < HACT) rpart += sn * areg[j];
+    if (j < HACT) {
+      float target = 0.6f * sinf(0.05f * t + 0.37f * (float)j);
+      float e = sn - target;
+      rpart += e * e;
+    }
   }
   if (lane < HACT) apart = areg[lane] * areg[lane];
   rpart = des::wave_reduce_sum_f(rpart);
   apart = des::wave_reduce_sum_f(apart);
   if (lane == 0) {
-    float reward = rpart / HACT - 0.05f * (apart / HACT);
+    float reward = 1.0f - rpart / HACT - 0.05f * (apart / HACT);
     if (alive[i]) { fitness[i] += reward; ep_len[i] += 1.0f; }
     int st = steps[i] + 1;
     if (st >= horizon) st = 0;
@@ -147,27 +163,30 @@ void des_launch_latent_reset(float* z, int* steps, float* fitness,
   hipLaunchKernelGGL(latent_reset_kernel, dim3(g), dim3(t), 0, stream, z,
                      steps, fitness, alive, ep_len, env_ids, seed, P);
 }
-void des_launch_latent_render(const float* z, uint16_t* obs, int C, int HW,
-                              int P, hipStream_t stream) {
+void des_launch_latent_render(const float* z, const float* coef,
+                              uint16_t* obs, int C, int HW, int P,
+                              hipStream_t stream) {
   int wpb = 4, g = (P + wpb - 1) / wpb;
   if (C == 4 && HW == 84)
     hipLaunchKernelGGL((latent_render_kernel<4, 84>), dim3(g), dim3(64 * wpb),
-                       0, stream, z, obs, P);
+                       0, stream, z, coef, obs, P);
   else if (C == 3 && HW == 64)
     hipLaunchKernelGGL((latent_render_kernel<3, 64>), dim3(g), dim3(64 * wpb),
-                       0, stream, z, obs, P);
+                       0, stream, z, coef, obs, P);
 }
-void des_launch_latent_step(float* z, int* steps, const int* actions,
-                            float* fitness, uint8_t* alive, float* ep_len,
-                            int num_actions, int horizon, int P,
+void des_launch_latent_step(float* z, const float* coef, int* steps,
+                            const int* actions, float* fitness, uint8_t* alive,
+                            float* ep_len, int num_actions, int horizon, int P,
                             hipStream_t stream) {
   int t = 256, g = (P + t - 1) / t;
   if (num_actions == 18)
     hipLaunchKernelGGL(latent_step_kernel<18>, dim3(g), dim3(t), 0, stream, z,
-                       steps, actions, fitness, alive, ep_len, horizon, P);
+                       coef, steps, actions, fitness, alive, ep_len, horizon,
+                       P);
   else
     hipLaunchKernelGGL(latent_step_kernel<15>, dim3(g), dim3(t), 0, stream, z,
-                       steps, actions, fitness, alive, ep_len, horizon, P);
+                       coef, steps, actions, fitness, alive, ep_len, horizon,
+                       P);
 }
 void des_launch_humanoid_reset(float* s, int* steps, float* fitness,
                                uint8_t* alive, float* ep_len,

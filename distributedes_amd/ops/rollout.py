@@ -135,6 +135,7 @@ class HipRollout:
                               for p in range(4)]
         elif e in ("atari57_synth", "procgen_synth"):
             self.sy_z = z(P, 8)
+            self.sy_coef = z(P, 6)       # per-env game coefficients
             self.sy_steps = z(P, dt=torch.int32)
             c, hw, _ = self.env.obs_shape
             self.obs_bf16 = z(P, c, hw, hw, dt=torch.bfloat16)
@@ -171,9 +172,14 @@ class HipRollout:
                                 self.fitness, self.alive, self.ep_len, ids,
                                 gen_seed)
         elif e in ("atari57_synth", "procgen_synth"):
+            # host-side single source of truth for the per-game coefficients
+            # (Atari-57 multi-task parameterization; classics for procgen)
+            from ..envs.synthetic import game_coefs
+            ng = self.env.n_games
+            self.sy_coef.copy_(game_coefs(ids % ng, ng))
             self.ext.latent_reset(self.sy_z, self.sy_steps, self.fitness,
                                   self.alive, self.ep_len, ids, gen_seed)
-            self.ext.latent_render(self.sy_z, self.obs_bf16)
+            self.ext.latent_render(self.sy_z, self.sy_coef, self.obs_bf16)
         elif e == "humanoid_synth":
             self.ext.humanoid_reset(self.hu_s, self.hu_steps, self.fitness,
                                     self.alive, self.ep_len, ids, gen_seed)
@@ -255,10 +261,11 @@ class HipRollout:
                                slot, self.actions_i, self.fitness, self.alive,
                                self.ep_len, self._ids, self.seed_buf)
         elif e in ("atari57_synth", "procgen_synth"):
-            self.ext.latent_step(self.sy_z, self.sy_steps, self.actions_i,
-                                 self.fitness, self.alive, self.ep_len,
-                                 self.env.num_actions, self.env.horizon)
-            self.ext.latent_render(self.sy_z, self.obs_bf16)
+            self.ext.latent_step(self.sy_z, self.sy_coef, self.sy_steps,
+                                 self.actions_i, self.fitness, self.alive,
+                                 self.ep_len, self.env.num_actions,
+                                 self.env.horizon)
+            self.ext.latent_render(self.sy_z, self.sy_coef, self.obs_bf16)
         elif e == "humanoid_synth":
             self.ext.humanoid_step(self.hu_s, self.hu_steps, self.actions_f,
                                    self.fitness, self.alive, self.ep_len,

@@ -41,11 +41,19 @@ class Comm:
         elif env_ws > 1:
             if backend is None:
                 backend = "nccl" if self.device.type == "cuda" else "gloo"
-            if "MASTER_ADDR" not in os.environ:
-                os.environ["MASTER_ADDR"] = "127.0.0.1"
-            if "MASTER_PORT" not in os.environ:
-                os.environ["MASTER_PORT"] = "29531"
-            dist.init_process_group(backend=backend)
+            store_file = os.environ.get("DES_FILE_STORE")
+            if store_file:
+                # test hook: file-based rendezvous (no TCP port to race on)
+                rank = int(os.environ.get("RANK", "0"))
+                store = dist.FileStore(store_file, env_ws)
+                dist.init_process_group(backend=backend, store=store,
+                                        rank=rank, world_size=env_ws)
+            else:
+                if "MASTER_ADDR" not in os.environ:
+                    os.environ["MASTER_ADDR"] = "127.0.0.1"
+                if "MASTER_PORT" not in os.environ:
+                    os.environ["MASTER_PORT"] = "29531"
+                dist.init_process_group(backend=backend)
             self.initialized_here = True
         self.world = dist.get_world_size() if dist.is_initialized() else 1
         self.rank = dist.get_rank() if dist.is_initialized() else 0

@@ -20,9 +20,9 @@ def _cfg(**kw):
     return preset("cartpole_es", **base)
 
 
-def _worker(rank, world, port, fn_name, out_q):
-    os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = str(port)
+def _worker(rank, world, store_file, fn_name, out_q):
+    # FileStore rendezvous via the Comm test hook: no TCP port to race on
+    os.environ["DES_FILE_STORE"] = store_file
     os.environ["WORLD_SIZE"] = str(world)
     os.environ["RANK"] = str(rank)
     os.environ["LOCAL_RANK"] = str(rank)
@@ -38,20 +38,12 @@ def _worker(rank, world, port, fn_name, out_q):
             dist.destroy_process_group()
 
 
-def _free_port():
-    import socket
-    sk = socket.socket()
-    sk.bind(("127.0.0.1", 0))
-    port = sk.getsockname()[1]
-    sk.close()
-    return port
-
-
-def _spawn(fn_name, world=2, port=None):
-    port = port or _free_port()
+def _spawn(fn_name, world=2):
+    import tempfile
+    store_file = os.path.join(tempfile.mkdtemp(prefix="des_store_"), "store")
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_worker, args=(r, world, port, fn_name, q))
+    procs = [ctx.Process(target=_worker, args=(r, world, store_file, fn_name, q))
              for r in range(world)]
     for p in procs:
         p.start()

@@ -24,11 +24,13 @@ def _cfg(**kw):
     return preset("cartpole_es", **base)
 
 
-def _init(rank, world, port):
-    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
-                      WORLD_SIZE=str(world), RANK=str(rank),
+def _init(rank, world, store_file):
+    os.environ.update(WORLD_SIZE=str(world), RANK=str(rank),
                       LOCAL_RANK=str(rank))
-    dist.init_process_group("gloo", rank=rank, world_size=world,
+    # FileStore rendezvous: no TCP port to race on (a bind-close-reuse
+    # _free_port() helper is a TOCTOU flake under a busy test box)
+    store = dist.FileStore(store_file, world)
+    dist.init_process_group("gloo", rank=rank, world_size=world, store=store,
                             timeout=datetime.timedelta(seconds=45))
 
 
@@ -73,24 +75,13 @@ def _resume_phase(rank, world, port, ckpt, out_q):
             dist.destroy_process_group()
 
 
-def _free_port():
-    import socket
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    port = s.getsockname()[1]
-    s.close()
-    return port
-
-
 @pytest.mark.timeout(180)
 def test_rank_death_then_checkpoint_resume(tmp_path):
     ckpt = str(tmp_path / "g2.pt")
-    _free_port._p1 = _free_port()
-    _free_port._p2 = _free_port()
-    _free_port._p3 = _free_port()
+    s1, s2, s3 = (str(tmp_path / f"store{i}") for i in (1, 2, 3))
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_crash_phase, args=(r, 2, _free_port._p1, ckpt, q))
+    procs = [ctx.Process(target=_crash_phase, args=(r, 2, s1, ckpt, q))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -104,7 +95,7 @@ def test_rank_death_then_checkpoint_resume(tmp_path):
 
     # fresh world resumes from the generation-2 checkpoint
     q2 = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_resume_phase, args=(r, 2, _free_port._p2, ckpt, q2))
+    procs = [ctx.Process(target=_resume_phase, args=(r, 2, s2, ckpt, q2))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -127,7 +118,7 @@ def test_rank_death_then_checkpoint_resume(tmp_path):
     # reassociation), so instead assert the resumed world matches a second
     # resumed world (determinism of recovery itself)
     q3 = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_resume_phase, args=(r, 2, _free_port._p3, ckpt, q3))
+    procs = [ctx.Process(target=_resume_phase, args=(r, 2, s3, ckpt, q3))
              for r in range(2)]
     for p in procs:
         p.start()

@@ -209,12 +209,14 @@ def test_pong_kernel_matches_torch_env(ext):
 
 def test_latent_and_humanoid_kernels_match(ext):
     from distributedes_amd.envs import Atari57SyntheticEnv, HumanoidSyntheticEnv
+    from distributedes_amd.envs.synthetic import game_coefs
     P = 8
     env = Atari57SyntheticEnv()
     env.reset(P, 5)
     dev = "cuda"
     ids = torch.arange(P, dtype=torch.int64, device=dev)
     zz = torch.zeros(P, 8, device=dev)
+    coef = game_coefs(ids % env.n_games, env.n_games).to(dev)
     steps = torch.zeros(P, dtype=torch.int32, device=dev)
     fit = torch.zeros(P, device=dev)
     alive = torch.ones(P, dtype=torch.uint8, device=dev)
@@ -222,12 +224,13 @@ def test_latent_and_humanoid_kernels_match(ext):
     obs = torch.zeros(P, 4, 84, 84, dtype=torch.bfloat16, device=dev)
     ext.latent_reset(zz, steps, fit, alive, eplen, ids, 5)
     torch.testing.assert_close(zz.cpu(), env.z, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(coef.cpu(), env.coef, rtol=0, atol=0)
     for t in range(5):
         a = torch.randint(0, 18, (P,))
         o_ref, r_ref, _ = env.step(a)
-        ext.latent_step(zz, steps, a.to(torch.int32).cuda(), fit, alive,
+        ext.latent_step(zz, coef, steps, a.to(torch.int32).cuda(), fit, alive,
                         eplen, 18, 1000)
-        ext.latent_render(zz, obs)
+        ext.latent_render(zz, coef, obs)
         torch.testing.assert_close(zz.cpu(), env.z, rtol=1e-3, atol=1e-4)
         torch.testing.assert_close(obs.cpu().float(), o_ref, rtol=0.05, atol=0.05)
 
