@@ -99,11 +99,18 @@ def test_nature_cnn_forward_logits_match_ref(ext):
     assert agree > 0.7, float(agree)
 
 
-def _nature_ref_logits(pol, w, obs):
+def _nature_ref_logits(pol, w, obs, bf16_acts=False):
+    """Grouped-conv fp32 reference.  bf16_acts=True additionally rounds the
+    inter-layer activations to bf16 — modeling the kernel's per-layer
+    storage dtype so each layer's rounding is budgeted explicitly (the
+    matched-dtype reference of the round-1 review)."""
     import torch.nn.functional as F
     P = w.shape[0]
     s = pol.spec
     x = obs.reshape(1, -1, 84, 84)
+
+    def rnd(t):
+        return t.to(torch.bfloat16).float() if bf16_acts else t
 
     def gconv(x, wn, bn, stride):
         W = s.slice(w, wn)
@@ -111,14 +118,90 @@ def _nature_ref_logits(pol, w, obs):
         Co = W.shape[1]
         return F.conv2d(x, W.reshape(P * Co, *W.shape[2:]), b.reshape(-1),
                         stride=stride, groups=P)
-    x = F.relu(gconv(x, "conv1.weight", "conv1.bias", 4))
-    x = F.relu(gconv(x, "conv2.weight", "conv2.bias", 2))
-    x = F.relu(gconv(x, "conv3.weight", "conv3.bias", 1))
+    x = rnd(F.relu(gconv(x, "conv1.weight", "conv1.bias", 4)))
+    x = rnd(F.relu(gconv(x, "conv2.weight", "conv2.bias", 2)))
+    x = rnd(F.relu(gconv(x, "conv3.weight", "conv3.bias", 1)))
     x = x.reshape(P, -1, 1)
-    x = F.relu(torch.baddbmm(s.slice(w, "fc.bias").unsqueeze(-1),
-                             s.slice(w, "fc.weight"), x))
+    x = rnd(F.relu(torch.baddbmm(s.slice(w, "fc.bias").unsqueeze(-1),
+                                 s.slice(w, "fc.weight"), x)))
     return torch.baddbmm(s.slice(w, "head.bias").unsqueeze(-1),
                          s.slice(w, "head.weight"), x).squeeze(-1)
+
+
+def _nature_fixed_inputs(P=8):
+    """Deterministic weights/frames for the golden + matched-dtype tests."""
+    g = torch.Generator().manual_seed(0xA57)
+    pol = NatureCNNPolicy(num_actions=18)
+    flat = (torch.randn(P, pol.dim, generator=g) * 0.05).to(torch.bfloat16)
+    obs_u8 = (torch.rand(P, 4, 84, 84, generator=g) < 0.15).to(torch.uint8)
+    return pol, flat, obs_u8
+
+
+def _nature_kernel_logits(ext, pol, flat, obs_u8):
+    P = flat.shape[0]
+    popw = _pad(flat.cuda())
+    slot_map = torch.tensor([0, 1, 2, 3], dtype=torch.int32).cuda()
+    acts = torch.zeros(P, dtype=torch.int32).cuda()
+    logits = torch.zeros(P, 18).cuda()
+    offs = torch.tensor(nature_cnn_offsets(pol.spec), dtype=torch.int64).cuda()
+    ext.nature_cnn_forward(popw, obs_u8.cuda(), slot_map, acts, logits, 18,
+                           offs)
+    return logits.cpu(), acts.cpu()
+
+
+def test_nature_cnn_matched_dtype_argmax(ext):
+    """Against the bf16-intermediates reference the kernel must agree on
+    >=90% of argmax actions and track logits within a per-layer bf16
+    error budget (round-1 review weak #3: 0.7 at fp32-ref is too loose)."""
+    pol, flat, obs_u8 = _nature_fixed_inputs(P=64)
+    logits, acts = _nature_kernel_logits(ext, pol, flat, obs_u8)
+    ref = _nature_ref_logits(pol, flat.float(), obs_u8.float(), bf16_acts=True)
+    torch.testing.assert_close(logits, ref, rtol=0.02, atol=0.02)
+    agree = (acts == ref.argmax(-1).to(torch.int32)).float().mean()
+    assert agree >= 0.9, float(agree)
+
+
+def test_nature_cnn_golden_logits(ext):
+    """Bit-exact regression pin: fixed-seed weights/frames vs a committed
+    golden tensor.  Catches 1-ulp-scale kernel regressions that tolerance
+    tests hide.  An INTENTIONAL numerics change must regenerate the golden:
+    delete tests/golden/nature_logits_v1.pt and rerun on a GPU box (the
+    test reseeds it under gpurun_out/golden/ for committing)."""
+    import os
+    pol, flat, obs_u8 = _nature_fixed_inputs(P=8)
+    logits, acts = _nature_kernel_logits(ext, pol, flat, obs_u8)
+    here = os.path.dirname(os.path.abspath(__file__))
+    gpath = os.path.join(here, "golden", "nature_logits_v1.pt")
+    if not os.path.exists(gpath):
+        outdir = os.path.join(here, "..", "gpurun_out", "golden")
+        os.makedirs(outdir, exist_ok=True)
+        torch.save({"logits": logits, "acts": acts},
+                   os.path.join(outdir, "nature_logits_v1.pt"))
+        pytest.skip("golden seeded under gpurun_out/golden/ - commit it to "
+                    "tests/golden/")
+    want = torch.load(gpath)
+    torch.testing.assert_close(logits, want["logits"], rtol=0, atol=0)
+    assert torch.equal(acts, want["acts"])
+
+
+def test_nature_cnn_pop_scale_action_distribution(ext):
+    """pop=8192-shape smoke: random-weight population must produce a healthy
+    action spread (no collapsed/NaN logits at bench scale)."""
+    torch.manual_seed(11)
+    pol = NatureCNNPolicy(num_actions=18)
+    P = 8192
+    flat = (torch.randn(P, pol.dim) * 0.05).to(torch.bfloat16).cuda()
+    obs_u8 = (torch.rand(P, 4, 84, 84) < 0.15).to(torch.uint8).cuda()
+    popw = _pad(flat)
+    slot_map = torch.tensor([0, 1, 2, 3], dtype=torch.int32).cuda()
+    acts = torch.zeros(P, dtype=torch.int32).cuda()
+    logits = torch.zeros(P, 18).cuda()
+    offs = torch.tensor(nature_cnn_offsets(pol.spec), dtype=torch.int64).cuda()
+    ext.nature_cnn_forward(popw, obs_u8, slot_map, acts, logits, 18, offs)
+    assert torch.isfinite(logits).all()
+    counts = torch.bincount(acts.to(torch.int64), minlength=18)
+    assert (counts > 0).sum() >= 12          # most actions represented
+    assert counts.max() < P * 0.5            # no single-action collapse
 
 
 def test_nature_cnn_slot_map_rolls_frames(ext):
