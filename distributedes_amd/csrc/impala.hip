@@ -18,11 +18,21 @@ namespace {
 typedef __attribute__((ext_vector_type(8))) short mfma_ab;
 typedef __attribute__((ext_vector_type(4))) float mfma_cd;
 
-// 3x3 pad-1 conv as MFMA implicit GEMM per spatial tile:
-//   M = TS*TS positions, N = CO, K = CI*9 (zero-padded to 32-multiples).
-// Per K-step all threads build the A-slab (im2col from the halo-staged
-// input tile) and B-slab (direct [co][k] rows of the member's weights),
-// then each wave runs one v_mfma_f32_16x16x32_bf16 per owned 16x16 C-tile.
+// 3x3 pad-1 conv as MFMA implicit GEMM per spatial tile, DIRECT-FRAGMENT
+// design (round-2 rewrite; the round-1 version staged an im2col A-slab per
+// K-step, and the committed ablation attributed 0.54 ms of the 0.81 ms
+// res-conv to that staging/barrier skeleton and 0.26 ms to the gather):
+//
+//   * the halo tile is staged TRANSPOSED: a_lds[y][x][ci] with ci
+//     contiguous (stride CIS, bank-spread).  The GEMM K order is then
+//     k' = tap*CIP + ci, so every MFMA A-fragment (8 consecutive k' for one
+//     tap) is ONE 16-byte ds_read_b128 at a shift-computed offset — no
+//     im2col pass, no A-slab, no re-read.
+//   * the member's weights are reordered once per block into the same
+//     k'-order (sb[co][k'], all K resident), so the k-loop contains ZERO
+//     barriers and zero staging: per K-step each wave issues one A-read +
+//     one B-read per owned 16x16 C-tile and the MFMA.
+//   M = TS*TS positions, N = CO, K' = CIP*9 zero-padded to 32-multiples.
 // Fragment maps hardware-verified by tools/probe_mfma.hip.
 template <int CI, int CO, int HW, int TS, bool RELU_IN, bool ADD_SKIP>
 __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
@@ -33,15 +43,22 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
     int64_t b_off, int P) {
   constexpr int TH = TS + 2;
   constexpr int M = TS * TS;               // 256 or 64
-  constexpr int K = CI * 9;
-  constexpr int KSTEPS = (K + 31) / 32;
+  constexpr int CIP = CI == 3 ? 8 : CI;    // ci zero-padded to an 8-multiple
+  constexpr int CIS = CIP == 32 ? 40 : 24; // LDS ci stride: 16B-aligned
+                                           // b128 reads, 16-way bank spread
+  constexpr int K = CIP * 9;               // k' = tap*CIP + ci
+  constexpr int KPAD = ((K + 31) / 32) * 32;
+  constexpr int KSTEPS = KPAD / 32;
+  constexpr int CPT = CIP / 8;             // 8-chunks per tap
+  constexpr int SBS = KPAD + 8;            // sb row stride (bank spread)
   constexpr int NWAVES = M / 64;           // 4 or 1
   constexpr int NTILES = (M / 16) * (CO / 16);
   constexpr int MAXT = NTILES / NWAVES;    // tiles per wave
-  __shared__ uint16_t a_lds[CI * TH * TH];
-  __shared__ uint16_t sa[M * 32];
-  __shared__ uint16_t sb[CO * 32];
-  __shared__ int tbl[32];  // k -> halo-tile offset (my=mx=0); -1 = zero-pad
+  constexpr int MP8 = M + 8;               // epilogue row stride (bank spread)
+  constexpr int ALDS = TH * TH * CIS;
+  constexpr int U0 = ALDS > CO * MP8 ? ALDS : CO * MP8;
+  __shared__ uint16_t u_lds[U0];           // halo, then epilogue staging
+  __shared__ uint16_t sb[CO * SBS];        // [co][k'] reordered weights
   const int i = blockIdx.x;
   if (i >= P) return;
   constexpr int NT = HW / TS;
@@ -49,76 +66,68 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
   const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
   const uint16_t* w = popw + (int64_t)i * row_stride;
 
-  // stage input halo tile (zero-padded at image edges, optional relu)
+  // stage the input halo tile transposed (zero-padded at image edges and in
+  // the ci pad lanes, optional fused relu); consecutive tids read
+  // consecutive x -> coalesced HBM
   const uint16_t* src = in + (int64_t)i * CI * HW * HW;
-  for (int e = tid; e < CI * TH * TH; e += M) {
+  for (int e = tid; e < CIP * TH * TH; e += M) {
     int ci = e / (TH * TH), r = e % (TH * TH);
-    int y = ty0 + r / TH - 1, x = tx0 + r % TH - 1;
+    int yl = r / TH, xl = r % TH;
+    int y = ty0 + yl - 1, x = tx0 + xl - 1;
     uint16_t v = 0;
-    if (y >= 0 && y < HW && x >= 0 && x < HW) {
+    if (ci < CI && y >= 0 && y < HW && x >= 0 && x < HW) {
       v = src[(ci * HW + y) * HW + x];
       if (RELU_IN && (v & 0x8000)) v = 0;
     }
-    a_lds[e] = v;
+    u_lds[(yl * TH + xl) * CIS + ci] = v;
+  }
+  // reorder weights [co][ci][tap] -> sb[co][tap*CIP + ci] (zero-fill pads);
+  // the 4 spatial blocks of a member hit the same 9-18 KiB region -> L2-warm
+  const uint16_t* wsrc = w + w_off;
+  for (int e = tid; e < CO * KPAD; e += M) {
+    int n = e / KPAD, kp = e % KPAD;
+    int tap = kp / CIP, ci = kp % CIP;
+    sb[n * SBS + kp] = (tap < 9 && ci < CI)
+                           ? wsrc[n * (CI * 9) + ci * 9 + tap]
+                           : (uint16_t)0;
   }
   __syncthreads();
 
-  const uint16_t* wsrc = w + w_off;
   mfma_cd acc[MAXT];
 #pragma unroll
   for (int u = 0; u < MAXT; ++u) acc[u] = 0.f;
+#pragma unroll 2
   for (int ks = 0; ks < KSTEPS; ++ks) {
-    // k -> offset table (separable: a_lds idx = tbl[kk] + my*TH + mx)
-    if (tid < 32) {
-      int k = ks * 32 + tid;
-      int off = -1;
-      if (k < K) {
-        int ci = k / 9, rm = k - ci * 9, ky = rm / 3, kx = rm - ky * 3;
-        off = (ci * TH + ky) * TH + kx;
-      }
-      tbl[tid] = off;
-    }
-    __syncthreads();
-    // A-slab [M][32]: im2col gather from the halo tile
-    for (int e = tid; e < M * 32 / 4; e += M) {  // 4 elems per unit
-      int m = (e * 4) >> 5, kk0 = (e * 4) & 31;
-      int mybase = (m / TS) * TH + (m % TS);
-      const int4 tq = *reinterpret_cast<const int4*>(tbl + kk0);
-      uint16_t v0 = tq.x < 0 ? 0 : a_lds[tq.x + mybase];
-      uint16_t v1 = tq.y < 0 ? 0 : a_lds[tq.y + mybase];
-      uint16_t v2 = tq.z < 0 ? 0 : a_lds[tq.z + mybase];
-      uint16_t v3 = tq.w < 0 ? 0 : a_lds[tq.w + mybase];
-      *reinterpret_cast<uint2*>(sa + m * 32 + kk0) =
-          uint2{(uint32_t)v0 | ((uint32_t)v1 << 16),
-                (uint32_t)v2 | ((uint32_t)v3 << 16)};
-    }
-    // B-slab [CO][32]: direct rows of the flat [co][k] weights
-    for (int e = tid; e < CO * 32; e += M) {
-      int n = e >> 5, kk = e & 31;
-      int k = ks * 32 + kk;
-      sb[n * 32 + kk] = k < K ? wsrc[n * K + k] : (uint16_t)0;
-    }
-    __syncthreads();
+    const int kc = ks * 4 + (lane >> 4);   // global 8-chunk index
+    int tap = kc / CPT;
+    if (tap > 8) tap = 8;                  // zero-pad taps: B rows are zero,
+    const int ci0 = (kc % CPT) * 8;        // A may read any in-bounds slot
+    const int ky = tap / 3, kx = tap - 3 * (tap / 3);
+    mfma_ab bfr[CO / 16];
+#pragma unroll
+    for (int nn = 0; nn < CO / 16; ++nn)
+      bfr[nn] = *reinterpret_cast<const mfma_ab*>(
+          sb + (nn * 16 + (lane & 15)) * SBS + ks * 32 + (lane >> 4) * 8);
 #pragma unroll
     for (int u = 0; u < MAXT; ++u) {
       int t = wid + NWAVES * u;
       int mt = t / (CO / 16), nt = t % (CO / 16);
+      int m = mt * 16 + (lane & 15);
+      int my = m / TS + ky, mx = m % TS + kx;
       mfma_ab af = *reinterpret_cast<const mfma_ab*>(
-          sa + (mt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
-      mfma_ab bf = *reinterpret_cast<const mfma_ab*>(
-          sb + (nt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
-      acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[u], 0, 0, 0);
+          u_lds + (my * TH + mx) * CIS + ci0);
+      acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr[nt], acc[u],
+                                                       0, 0, 0);
     }
-    __syncthreads();
   }
+  __syncthreads();   // u_lds is re-carved as the epilogue staging buffer
 
   // epilogue: the naive per-lane scatter hits a 2*HW*HW-byte stride (one
-  // cache line per lane) — stage the bias+relu'd tile in LDS (sa is free
-  // and exactly CO*M elements) and do a coalesced cooperative store instead
+  // cache line per lane) — stage the bias'd tile in LDS (row stride M+8 to
+  // spread banks) and do a coalesced cooperative uint4 store instead
   const uint16_t* bias = w + b_off;
   uint16_t* dst = out + (int64_t)i * CO * HW * HW;
   const uint16_t* sk = ADD_SKIP ? skip + (int64_t)i * CO * HW * HW : nullptr;
-  static_assert(CO * M <= M * 32, "output tile exceeds the sa staging carve");
 #pragma unroll
   for (int u = 0; u < MAXT; ++u) {
     int t = wid + NWAVES * u;
@@ -128,17 +137,16 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
     float b = des::bf16_to_f32(bias[n]);
 #pragma unroll
     for (int r = 0; r < 4; ++r)
-      sa[n * M + m0 + r] = des::f32_to_bf16(acc[u][r] + b);
+      u_lds[n * MP8 + m0 + r] = des::f32_to_bf16(acc[u][r] + b);
   }
   __syncthreads();
   // 8 consecutive u16 of a tile row are contiguous + 16B-aligned in dst
   // (tx0 is a multiple of TS >= 8): store uint4 per thread (ablation showed
   // the narrow store tail at ~350 GB/s was 23% of the kernel)
   for (int e8 = tid; e8 < CO * M / 8; e8 += M) {
-    int e = e8 * 8;
-    int n = e / M, m = e - n * M;
+    int n = e8 / (M / 8), m = (e8 % (M / 8)) * 8;
     int64_t gidx = ((int64_t)n * HW + ty0 + m / TS) * HW + tx0 + m % TS;
-    uint4 v = *reinterpret_cast<const uint4*>(sa + e);
+    uint4 v = *reinterpret_cast<const uint4*>(u_lds + n * MP8 + m);
     if (ADD_SKIP) {
       uint4 sv = *reinterpret_cast<const uint4*>(sk + gidx);
       const uint16_t* a = reinterpret_cast<const uint16_t*>(&v);
@@ -156,24 +164,47 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
 template <int C, int HW>  // output is HW/2 x HW/2 (3x3 s2 pad1)
 __global__ void impala_pool_kernel(const uint16_t* __restrict__ in,
                                    uint16_t* __restrict__ out, int P) {
+  // 8 contiguous outputs per thread via uint4 row loads (the round-1
+  // one-scalar-per-output version issued 9 scalar loads/output and ran
+  // ~15x over its HBM floor)
   constexpr int HO = HW / 2;
-  int64_t n = (int64_t)P * C * HO * HO;
+  constexpr int GRP = HO / 8;              // 8-output groups per out row
+  int64_t nunits = (int64_t)P * C * HO * GRP;
   int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (idx >= n) return;
-  int ox = idx % HO, oy = (idx / HO) % HO;
-  int64_t pc = idx / (HO * HO);  // member*C + c
+  if (idx >= nunits) return;
+  int g = (int)(idx % GRP);
+  int oy = (int)((idx / GRP) % HO);
+  int64_t pc = idx / ((int64_t)GRP * HO);  // member*C + c
   const uint16_t* src = in + pc * HW * HW;
-  float m = -3.0e38f;
+  const int xb = 2 * (g * 8);              // 16B-aligned row base
+  float best[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) best[j] = -3.0e38f;
   for (int dy = -1; dy <= 1; ++dy) {
     int y = 2 * oy + dy;
     if (y < 0 || y >= HW) continue;
-    for (int dx = -1; dx <= 1; ++dx) {
-      int x = 2 * ox + dx;
-      if (x < 0 || x >= HW) continue;
-      m = fmaxf(m, des::bf16_to_f32(src[y * HW + x]));
+    const uint16_t* row = src + y * HW;
+    uint4 q0 = *reinterpret_cast<const uint4*>(row + xb);
+    uint4 q1 = *reinterpret_cast<const uint4*>(row + xb + 8);
+    float e[17];
+    e[0] = xb > 0 ? des::bf16_to_f32(row[xb - 1]) : -3.0e38f;
+    const uint16_t* h0 = reinterpret_cast<const uint16_t*>(&q0);
+    const uint16_t* h1 = reinterpret_cast<const uint16_t*>(&q1);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      e[1 + j] = des::bf16_to_f32(h0[j]);
+      e[9 + j] = des::bf16_to_f32(h1[j]);
     }
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      best[j] = fmaxf(best[j],
+                      fmaxf(e[2 * j], fmaxf(e[2 * j + 1], e[2 * j + 2])));
   }
-  out[idx] = des::f32_to_bf16(m);
+  uint16_t o[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) o[j] = des::f32_to_bf16(best[j]);
+  *reinterpret_cast<uint4*>(out + pc * HO * HO + oy * HO + g * 8) =
+      *reinterpret_cast<const uint4*>(o);
 }
 
 // relu-flatten 2048 -> fc 256 relu -> head A + argmax
@@ -286,7 +317,7 @@ void des_launch_impala_conv3(const uint16_t* in, uint16_t* out,
 
 void des_launch_impala_pool(const uint16_t* in, uint16_t* out, int C, int HW,
                             int P, hipStream_t stream) {
-  int64_t n = (int64_t)P * C * (HW / 2) * (HW / 2);
+  int64_t n = (int64_t)P * C * (HW / 2) * ((HW / 2) / 8);  // 8 outputs/thread
   int g = (int)((n + 255) / 256);
   if (C == 16 && HW == 64)
     hipLaunchKernelGGL((impala_pool_kernel<16, 64>), dim3(g), dim3(256), 0,
