@@ -291,6 +291,10 @@ void impala_conv(torch::Tensor in, torch::Tensor out, torch::Tensor skip,
   CHECK_DEV(in); CHECK_CONT(in); CHECK_DEV(out); CHECK_CONT(out);
   CHECK_DEV(popw); CHECK_CONT(popw);
   TORCH_CHECK(in.dtype() == torch::kBFloat16 && out.dtype() == torch::kBFloat16);
+  // the kernel's vectorized weight reorder does uint4 (8-elem) loads from
+  // popw+row*stride+w_off: both strides must be 8-elem aligned
+  TORCH_CHECK(w_off % 8 == 0 && popw.size(1) % 8 == 0,
+              "impala_conv: weight offset/stride must be 8-elem aligned");
   int P = popw.size(0);
   const uint16_t* sk = add_skip ? u16p(skip) : nullptr;
   des_launch_impala_conv3(u16p(in), (uint16_t*)out.data_ptr(), sk, u16p(popw),

@@ -34,17 +34,21 @@ typedef __attribute__((ext_vector_type(4))) float mfma_cd;
 //     one B-read per owned 16x16 C-tile and the MFMA.
 //   M = TS*TS positions, N = CO, K' = CIP*9 zero-padded to 32-multiples.
 // Fragment maps hardware-verified by tools/probe_mfma.hip.
-template <int CI, int CO, int HW, int TS, bool RELU_IN, bool ADD_SKIP>
-__global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
+template <int CI, int CO, int HW, int TSY, int TSX, bool RELU_IN,
+          bool ADD_SKIP>
+__global__ __launch_bounds__(TSY * TSX) void impala_conv3_kernel(
     const uint16_t* __restrict__ in,    // [P][CI][HW][HW]
     uint16_t* __restrict__ out,         // [P][CO][HW][HW]
     const uint16_t* __restrict__ skip,  // nullable [P][CO][HW][HW]
     const uint16_t* __restrict__ popw, int64_t row_stride, int64_t w_off,
     int64_t b_off, int P) {
-  constexpr int TH = TS + 2;
-  constexpr int M = TS * TS;               // 256 or 64
+  constexpr int THY = TSY + 2, THX = TSX + 2;
+  constexpr int M = TSY * TSX;             // 64..512 (rectangular tiles:
+                                           // wider tiles amortize the weight
+                                           // reorder and cut the halo ratio)
   constexpr int CIP = CI == 3 ? 8 : CI;    // ci zero-padded to an 8-multiple
-  constexpr int CIS = CIP == 32 ? 40 : 24; // LDS ci stride: 16B-aligned
+  constexpr int CIS = CIP == 8 ? 8 : (CIP == 16 ? 24 : 40);
+                                           // LDS ci stride: 16B-aligned
                                            // b128 reads, 16-way bank spread
   constexpr int K = CIP * 9;               // k' = tap*CIP + ci
   constexpr int KPAD = ((K + 31) / 32) * 32;
@@ -55,41 +59,63 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
   constexpr int NTILES = (M / 16) * (CO / 16);
   constexpr int MAXT = NTILES / NWAVES;    // tiles per wave
   constexpr int MP8 = M + 8;               // epilogue row stride (bank spread)
-  constexpr int ALDS = TH * TH * CIS;
+  constexpr int ALDS = THY * THX * CIS;
   constexpr int U0 = ALDS > CO * MP8 ? ALDS : CO * MP8;
   __shared__ uint16_t u_lds[U0];           // halo, then epilogue staging
   __shared__ uint16_t sb[CO * SBS];        // [co][k'] reordered weights
   const int i = blockIdx.x;
   if (i >= P) return;
-  constexpr int NT = HW / TS;
-  const int ty0 = (blockIdx.y / NT) * TS, tx0 = (blockIdx.y % NT) * TS;
+  constexpr int NTX = HW / TSX;
+  const int ty0 = (blockIdx.y / NTX) * TSY, tx0 = (blockIdx.y % NTX) * TSX;
   const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
   const uint16_t* w = popw + (int64_t)i * row_stride;
 
   // stage the input halo tile transposed (zero-padded at image edges and in
   // the ci pad lanes, optional fused relu); consecutive tids read
   // consecutive x -> coalesced HBM
+  // the CIP pad lanes must be ZEROED, not left as garbage: their sb rows
+  // are zero, but uninitialized LDS can hold NaN bit patterns and
+  // NaN * 0 = NaN would poison the MFMA accumulator
+  if constexpr (CIP > CI) {
+    for (int e = tid; e < (CIP - CI) * THY * THX; e += M) {
+      int ci = CI + e / (THY * THX), r = e % (THY * THX);
+      u_lds[r * CIS + ci] = 0;
+    }
+  }
   const uint16_t* src = in + (int64_t)i * CI * HW * HW;
-  for (int e = tid; e < CIP * TH * TH; e += M) {
-    int ci = e / (TH * TH), r = e % (TH * TH);
-    int yl = r / TH, xl = r % TH;
+  for (int e = tid; e < CI * THY * THX; e += M) {
+    int ci = e / (THY * THX), r = e % (THY * THX);
+    int yl = r / THX, xl = r % THX;
     int y = ty0 + yl - 1, x = tx0 + xl - 1;
     uint16_t v = 0;
-    if (ci < CI && y >= 0 && y < HW && x >= 0 && x < HW) {
+    if (y >= 0 && y < HW && x >= 0 && x < HW) {
       v = src[(ci * HW + y) * HW + x];
       if (RELU_IN && (v & 0x8000)) v = 0;
     }
-    u_lds[(yl * TH + xl) * CIS + ci] = v;
+    u_lds[(yl * THX + xl) * CIS + ci] = v;
   }
-  // reorder weights [co][ci][tap] -> sb[co][tap*CIP + ci] (zero-fill pads);
-  // the 4 spatial blocks of a member hit the same 9-18 KiB region -> L2-warm
+  // reorder weights [co][ci][tap] -> sb[co][tap*CIP + ci]: uint4 reads of
+  // the contiguous [co][k] source (coalesced HBM; w_off is 8-elem aligned,
+  // checked in bindings) scattered into the tap-major LDS layout
   const uint16_t* wsrc = w + w_off;
-  for (int e = tid; e < CO * KPAD; e += M) {
-    int n = e / KPAD, kp = e % KPAD;
-    int tap = kp / CIP, ci = kp % CIP;
-    sb[n * SBS + kp] = (tap < 9 && ci < CI)
-                           ? wsrc[n * (CI * 9) + ci * 9 + tap]
-                           : (uint16_t)0;
+  constexpr int KSRC = CI * 9;
+  if constexpr (KPAD > KSRC || CIP > CI) {   // zero the pad slots
+    for (int e = tid; e < CO * KPAD; e += M) {
+      int kp = e % KPAD;
+      int tap = kp / CIP, ci = kp % CIP;
+      if (tap >= 9 || ci >= CI) sb[(e / KPAD) * SBS + kp] = 0;
+    }
+  }
+  for (int e8 = tid; e8 < CO * KSRC / 8; e8 += M) {
+    uint4 q = *reinterpret_cast<const uint4*>(wsrc + e8 * 8);
+    const uint16_t* h = reinterpret_cast<const uint16_t*>(&q);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int idx = e8 * 8 + j;
+      int n = idx / KSRC, k = idx - n * KSRC;
+      int ci = k / 9, tap = k - 9 * ci;
+      sb[n * SBS + tap * CIP + ci] = h[j];
+    }
   }
   __syncthreads();
 
@@ -113,9 +139,9 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
       int t = wid + NWAVES * u;
       int mt = t / (CO / 16), nt = t % (CO / 16);
       int m = mt * 16 + (lane & 15);
-      int my = m / TS + ky, mx = m % TS + kx;
+      int my = m / TSX + ky, mx = m % TSX + kx;
       mfma_ab af = *reinterpret_cast<const mfma_ab*>(
-          u_lds + (my * TH + mx) * CIS + ci0);
+          u_lds + (my * THX + mx) * CIS + ci0);
       acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr[nt], acc[u],
                                                        0, 0, 0);
     }
@@ -145,7 +171,7 @@ __global__ __launch_bounds__(TS * TS) void impala_conv3_kernel(
   // the narrow store tail at ~350 GB/s was 23% of the kernel)
   for (int e8 = tid; e8 < CO * M / 8; e8 += M) {
     int n = e8 / (M / 8), m = (e8 % (M / 8)) * 8;
-    int64_t gidx = ((int64_t)n * HW + ty0 + m / TS) * HW + tx0 + m % TS;
+    int64_t gidx = ((int64_t)n * HW + ty0 + m / TSX) * HW + tx0 + m % TSX;
     uint4 v = *reinterpret_cast<const uint4*>(u_lds + n * MP8 + m);
     if (ADD_SKIP) {
       uint4 sv = *reinterpret_cast<const uint4*>(sk + gidx);
@@ -290,28 +316,31 @@ void des_launch_impala_conv3(const uint16_t* in, uint16_t* out,
                              int64_t row_stride, int64_t w_off, int64_t b_off,
                              int shape, bool relu_in, bool add_skip, int P,
                              hipStream_t stream) {
-#define CASE(S, CI, CO, HW, TS)                                               \
+#define CASE(S, CI, CO, HW, TSY, TSX)                                         \
   if (shape == S) {                                                           \
-    dim3 grid((uint32_t)P, (HW / TS) * (HW / TS));                            \
+    dim3 grid((uint32_t)P, (HW / TSY) * (HW / TSX));                          \
     if (relu_in && add_skip)                                                  \
-      hipLaunchKernelGGL((impala_conv3_kernel<CI, CO, HW, TS, true, true>),   \
-                         grid, dim3(TS * TS), 0, stream, in, out, skip, popw, \
-                         row_stride, w_off, b_off, P);                        \
+      hipLaunchKernelGGL(                                                     \
+          (impala_conv3_kernel<CI, CO, HW, TSY, TSX, true, true>), grid,      \
+          dim3(TSY * TSX), 0, stream, in, out, skip, popw, row_stride, w_off, \
+          b_off, P);                                                          \
     else if (relu_in)                                                         \
-      hipLaunchKernelGGL((impala_conv3_kernel<CI, CO, HW, TS, true, false>),  \
-                         grid, dim3(TS * TS), 0, stream, in, out, skip, popw, \
-                         row_stride, w_off, b_off, P);                        \
+      hipLaunchKernelGGL(                                                     \
+          (impala_conv3_kernel<CI, CO, HW, TSY, TSX, true, false>), grid,     \
+          dim3(TSY * TSX), 0, stream, in, out, skip, popw, row_stride, w_off, \
+          b_off, P);                                                          \
     else                                                                      \
-      hipLaunchKernelGGL((impala_conv3_kernel<CI, CO, HW, TS, false, false>), \
-                         grid, dim3(TS * TS), 0, stream, in, out, skip, popw, \
-                         row_stride, w_off, b_off, P);                        \
+      hipLaunchKernelGGL(                                                     \
+          (impala_conv3_kernel<CI, CO, HW, TSY, TSX, false, false>), grid,    \
+          dim3(TSY * TSX), 0, stream, in, out, skip, popw, row_stride, w_off, \
+          b_off, P);                                                          \
     return;                                                                   \
   }
-  CASE(0, 3, 16, 64, 16)
-  CASE(1, 16, 16, 32, 16)
-  CASE(2, 16, 32, 32, 16)
-  CASE(3, 32, 32, 16, 16)
-  CASE(4, 32, 32, 8, 8)
+  CASE(0, 3, 16, 64, 16, 32)
+  CASE(1, 16, 16, 32, 16, 32)
+  CASE(2, 16, 32, 32, 16, 32)
+  CASE(3, 32, 32, 16, 16, 16)
+  CASE(4, 32, 32, 8, 8, 8)
 #undef CASE
 }
 
