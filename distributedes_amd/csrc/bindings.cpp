@@ -53,6 +53,9 @@ void des_launch_centered_rank(const float*, float*, int, hipStream_t);
 void des_launch_impala_conv3(const uint16_t*, uint16_t*, const uint16_t*,
                              const uint16_t*, int64_t, int64_t, int64_t, int,
                              bool, bool, int, hipStream_t);
+void des_launch_impala_resblock(uint16_t*, const uint16_t*, int64_t, int64_t,
+                                int64_t, int64_t, int64_t, int, int,
+                                hipStream_t);
 void des_launch_impala_pool(const uint16_t*, uint16_t*, int, int, int,
                             hipStream_t);
 void des_launch_impala_fc_head(const uint16_t*, int64_t, const uint16_t*,
@@ -302,6 +305,20 @@ void impala_conv(torch::Tensor in, torch::Tensor out, torch::Tensor skip,
                           add_skip, P, cur_stream());
 }
 
+void impala_resblock(torch::Tensor x, torch::Tensor popw, int64_t w1_off,
+                     int64_t b1_off, int64_t w2_off, int64_t b2_off) {
+  CHECK_DEV(x); CHECK_CONT(x); CHECK_DEV(popw); CHECK_CONT(popw);
+  TORCH_CHECK(x.dtype() == torch::kBFloat16);
+  TORCH_CHECK(x.size(1) == 32, "resblock kernel is C=32 only");
+  TORCH_CHECK(w1_off % 8 == 0 && w2_off % 8 == 0 && popw.size(1) % 8 == 0,
+              "impala_resblock: weight offsets/stride must be 8-elem aligned");
+  int HW = x.size(2);
+  TORCH_CHECK(HW == 16 || HW == 8, "resblock kernel needs whole-image tiles");
+  des_launch_impala_resblock((uint16_t*)x.data_ptr(), u16p(popw),
+                             popw.size(1), w1_off, b1_off, w2_off, b2_off, HW,
+                             popw.size(0), cur_stream());
+}
+
 void impala_pool(torch::Tensor in, torch::Tensor out, int64_t C, int64_t HW) {
   CHECK_DEV(in); CHECK_CONT(in); CHECK_DEV(out); CHECK_CONT(out);
   int P = in.size(0);
@@ -360,6 +377,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("normalizer_observe", &normalizer_observe);
   m.def("centered_rank", &centered_rank, "K9: stable centered-rank weights");
   m.def("impala_conv", &impala_conv, "K6: 3x3 pad-1 conv (pop, tiled)");
+  m.def("impala_resblock", &impala_resblock,
+        "K6: fused residual block x += conv2(relu(conv1(relu(x)))), HW<=16");
   m.def("impala_pool", &impala_pool, "K6: 3x3 s2 pad-1 max pool");
   m.def("impala_fc_head", &impala_fc_head, "K6: fc 2048->256 + head + argmax");
 }

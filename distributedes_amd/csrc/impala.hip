@@ -187,6 +187,156 @@ __global__ __launch_bounds__(TSY * TSX) void impala_conv3_kernel(
   }
 }
 
+// Fused residual block for whole-image tiles (HW <= 16, C = 32):
+//   x <- x + conv2(relu(conv1(relu(x))))        (in place)
+// With the tile covering the whole image, conv1's output ring IS conv2's
+// zero padding, so y1 never touches HBM: after conv1's k-loop (barrier) the
+// relu'd result is written back into the SAME transposed LDS tile the
+// relu'd input occupied, and conv2 runs from it.  vs two conv3 launches
+// this removes the intermediate tensor's HBM write+read, one halo stage,
+// one weight-reorder barrier sequence and one kernel launch per pair.
+template <int C, int HW>
+__global__ __launch_bounds__(HW * HW) void impala_resblock_kernel(
+    uint16_t* __restrict__ x,            // [P][C][HW][HW], in & out
+    const uint16_t* __restrict__ popw, int64_t row_stride, int64_t w1_off,
+    int64_t b1_off, int64_t w2_off, int64_t b2_off, int P) {
+  constexpr int TH = HW + 2;
+  constexpr int M = HW * HW;        // 256 or 64
+  constexpr int CIS = 40;           // C == 32: aligned b128, bank-spread
+  constexpr int K = C * 9;          // 288 (= KPAD: no pad taps)
+  constexpr int KSTEPS = K / 32;    // 9; one tap per K-step
+  constexpr int SBS = K + 8;
+  constexpr int NWAVES = M / 64;
+  constexpr int NTILES = (M / 16) * (C / 16);
+  constexpr int MAXT = NTILES / NWAVES;
+  constexpr int MP8 = M + 8;
+  constexpr int RING = 4 * TH - 4;  // zero-padding ring positions
+  constexpr int ALDS = TH * TH * CIS;
+  static_assert(ALDS >= C * MP8, "epilogue staging must fit the tile buf");
+  __shared__ uint16_t ax[ALDS];     // relu(x) -> relu(y1) -> epilogue stage
+  __shared__ uint16_t sb[C * SBS];  // w1 then w2, tap-major k' order
+  const int i = blockIdx.x;
+  if (i >= P) return;
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  const uint16_t* w = popw + (int64_t)i * row_stride;
+  uint16_t* xme = x + (int64_t)i * C * M;
+
+  // zero the ring once (both convs consume it as zero padding)
+  for (int e = tid; e < RING * C; e += M) {
+    int rp = e / C, ci = e % C;
+    int yl, xl;
+    if (rp < TH) { yl = 0; xl = rp; }
+    else if (rp < 2 * TH) { yl = TH - 1; xl = rp - TH; }
+    else { int rr = rp - 2 * TH; yl = 1 + (rr >> 1); xl = (rr & 1) ? TH - 1 : 0; }
+    ax[(yl * TH + xl) * CIS + ci] = 0;
+  }
+  // stage relu(x) interior, transposed [y][x][ci]
+  for (int e = tid; e < C * M; e += M) {
+    int ci = e / M, r = e - ci * M;
+    uint16_t v = xme[e];
+    if (v & 0x8000) v = 0;
+    ax[((r / HW + 1) * TH + (r % HW + 1)) * CIS + ci] = v;
+  }
+  auto stage_sb = [&](const uint16_t* wsrc) {
+    for (int e8 = tid; e8 < C * K / 8; e8 += M) {
+      uint4 q = *reinterpret_cast<const uint4*>(wsrc + e8 * 8);
+      const uint16_t* h = reinterpret_cast<const uint16_t*>(&q);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int idx = e8 * 8 + j;
+        int n = idx / K, k = idx - n * K;
+        int ci = k / 9, tap = k - 9 * ci;
+        sb[n * SBS + tap * C + ci] = h[j];
+      }
+    }
+  };
+  auto run_conv = [&](mfma_cd* acc) {
+#pragma unroll 2
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      const int ky = ks / 3, kx = ks - 3 * (ks / 3);  // tap == ks for C=32
+      const int ci0 = (lane >> 4) * 8;
+      mfma_ab bfr[C / 16];
+#pragma unroll
+      for (int nn = 0; nn < C / 16; ++nn)
+        bfr[nn] = *reinterpret_cast<const mfma_ab*>(
+            sb + (nn * 16 + (lane & 15)) * SBS + ks * 32 + (lane >> 4) * 8);
+#pragma unroll
+      for (int u = 0; u < MAXT; ++u) {
+        int t = wid + NWAVES * u;
+        int mt = t / (C / 16), nt = t % (C / 16);
+        int m = mt * 16 + (lane & 15);
+        int my = m / HW + ky, mx = m % HW + kx;
+        mfma_ab af = *reinterpret_cast<const mfma_ab*>(
+            ax + (my * TH + mx) * CIS + ci0);
+        acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr[nt], acc[u],
+                                                         0, 0, 0);
+      }
+    }
+  };
+  stage_sb(w + w1_off);
+  __syncthreads();
+
+  mfma_cd acc[MAXT];
+#pragma unroll
+  for (int u = 0; u < MAXT; ++u) acc[u] = 0.f;
+  run_conv(acc);
+  __syncthreads();  // conv1 A/B reads complete
+
+  // y1 = relu(acc + b1) back into ax interior; w2 into sb
+  const uint16_t* b1 = w + b1_off;
+#pragma unroll
+  for (int u = 0; u < MAXT; ++u) {
+    int t = wid + NWAVES * u;
+    int mt = t / (C / 16), nt = t % (C / 16);
+    int n = nt * 16 + (lane & 15);
+    int m0 = mt * 16 + ((lane >> 4) << 2);
+    float b = des::bf16_to_f32(b1[n]);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float v = acc[u][r] + b;
+      int m = m0 + r;
+      ax[((m / HW + 1) * TH + (m % HW + 1)) * CIS + n] =
+          des::f32_to_bf16(v > 0.f ? v : 0.f);
+    }
+  }
+  stage_sb(w + w2_off);
+  __syncthreads();
+
+#pragma unroll
+  for (int u = 0; u < MAXT; ++u) acc[u] = 0.f;
+  run_conv(acc);
+  __syncthreads();  // conv2 reads complete; ax becomes the epilogue stage
+
+  const uint16_t* b2 = w + b2_off;
+#pragma unroll
+  for (int u = 0; u < MAXT; ++u) {
+    int t = wid + NWAVES * u;
+    int mt = t / (C / 16), nt = t % (C / 16);
+    int n = nt * 16 + (lane & 15);
+    int m0 = mt * 16 + ((lane >> 4) << 2);
+    float b = des::bf16_to_f32(b2[n]);
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      ax[n * MP8 + m0 + r] = des::f32_to_bf16(acc[u][r] + b);
+  }
+  __syncthreads();
+  // in-place skip-add store: each thread reads its x chunk then overwrites
+  // it (whole-image tile: this block is the only reader of member i's x)
+  for (int e8 = tid; e8 < C * M / 8; e8 += M) {
+    int n = e8 / (M / 8), m = (e8 - n * (M / 8)) * 8;
+    int64_t gidx = (int64_t)n * M + m;
+    uint4 v = *reinterpret_cast<const uint4*>(ax + n * MP8 + m);
+    uint4 sv = *reinterpret_cast<const uint4*>(xme + gidx);
+    const uint16_t* a = reinterpret_cast<const uint16_t*>(&v);
+    const uint16_t* b = reinterpret_cast<const uint16_t*>(&sv);
+    uint16_t o[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = des::f32_to_bf16(des::bf16_to_f32(a[j]) + des::bf16_to_f32(b[j]));
+    *reinterpret_cast<uint4*>(xme + gidx) = *reinterpret_cast<const uint4*>(o);
+  }
+}
+
 template <int C, int HW>  // output is HW/2 x HW/2 (3x3 s2 pad1)
 __global__ void impala_pool_kernel(const uint16_t* __restrict__ in,
                                    uint16_t* __restrict__ out, int P) {
@@ -342,6 +492,20 @@ void des_launch_impala_conv3(const uint16_t* in, uint16_t* out,
   CASE(3, 32, 32, 16, 16, 16)
   CASE(4, 32, 32, 8, 8, 8)
 #undef CASE
+}
+
+void des_launch_impala_resblock(uint16_t* x, const uint16_t* popw,
+                                int64_t row_stride, int64_t w1_off,
+                                int64_t b1_off, int64_t w2_off, int64_t b2_off,
+                                int HW, int P, hipStream_t stream) {
+  if (HW == 16)
+    hipLaunchKernelGGL((impala_resblock_kernel<32, 16>), dim3((uint32_t)P),
+                       dim3(256), 0, stream, x, popw, row_stride, w1_off,
+                       b1_off, w2_off, b2_off, P);
+  else if (HW == 8)
+    hipLaunchKernelGGL((impala_resblock_kernel<32, 8>), dim3((uint32_t)P),
+                       dim3(64), 0, stream, x, popw, row_stride, w1_off,
+                       b1_off, w2_off, b2_off, P);
 }
 
 void des_launch_impala_pool(const uint16_t* in, uint16_t* out, int C, int HW,

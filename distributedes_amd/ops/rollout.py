@@ -148,10 +148,8 @@ class HipRollout:
             self.it32 = z(P, 16, 32, 32, dt=bf)
             self.ib32 = z(P, 32, 32, 32, dt=bf)
             self.ix16 = z(P, 32, 16, 16, dt=bf)
-            self.it16 = z(P, 32, 16, 16, dt=bf)
             self.ib16 = z(P, 32, 16, 16, dt=bf)
             self.ix8 = z(P, 32, 8, 8, dt=bf)
-            self.it8 = z(P, 32, 8, 8, dt=bf)
         elif e == "humanoid_synth":
             self.hu_s = z(P, 376)
             self.hu_steps = z(P, dt=torch.int32)
@@ -231,18 +229,19 @@ class HipRollout:
         for r in (1, 2):
             conv(self.ix32, self.it32, None, f"s1r{r}c1", 1, True, False)
             conv(self.it32, self.ix32, self.ix32, f"s1r{r}c2", 1, True, True)
-        # stage 2
+        # stages 2/3: whole-image tiles (HW <= 16) -> fused residual pairs
+        # (y1 lives in LDS; no intermediate tensor, half the launches)
         conv(self.ix32, self.ib32, None, "s2", 2, False, False)
         ext.impala_pool(self.ib32, self.ix16, 32, 32)
         for r in (1, 2):
-            conv(self.ix16, self.it16, None, f"s2r{r}c1", 3, True, False)
-            conv(self.it16, self.ix16, self.ix16, f"s2r{r}c2", 3, True, True)
+            c1, c2 = io[f"s2r{r}c1"], io[f"s2r{r}c2"]
+            ext.impala_resblock(self.ix16, pop_w, c1[0], c1[1], c2[0], c2[1])
         # stage 3
         conv(self.ix16, self.ib16, None, "s3", 3, False, False)
         ext.impala_pool(self.ib16, self.ix8, 32, 16)
         for r in (1, 2):
-            conv(self.ix8, self.it8, None, f"s3r{r}c1", 4, True, False)
-            conv(self.it8, self.ix8, self.ix8, f"s3r{r}c2", 4, True, True)
+            c1, c2 = io[f"s3r{r}c1"], io[f"s3r{r}c2"]
+            ext.impala_resblock(self.ix8, pop_w, c1[0], c1[1], c2[0], c2[1])
         ext.impala_fc_head(pop_w, self.ix8.view(self.P, -1), self.actions_i,
                            logits if logits is not None else E,
                            self.env.num_actions, io["fc"][0], io["fc"][1],
