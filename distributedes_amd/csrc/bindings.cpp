@@ -47,8 +47,10 @@ void des_launch_humanoid_reset(float*, int*, float*, uint8_t*, float*,
                                const int64_t*, int64_t, int, hipStream_t);
 void des_launch_humanoid_step(float*, int*, const float*, float*, uint8_t*,
                               float*, int, int, hipStream_t);
-void des_launch_normalizer_observe(const float*, double*, double*, int, int,
+void des_launch_normalizer_observe(const float*, float*, int, int,
                                    hipStream_t);
+void des_launch_normalizer_observe_flush(const float*, double*, double*, int,
+                                         int, hipStream_t);
 void des_launch_centered_rank(const float*, float*, int, hipStream_t);
 void des_launch_impala_conv3(const uint16_t*, uint16_t*, const uint16_t*,
                              const uint16_t*, int64_t, int64_t, int64_t, int,
@@ -346,13 +348,23 @@ void centered_rank(torch::Tensor fitness, torch::Tensor u) {
                            cur_stream());
 }
 
-void normalizer_observe(torch::Tensor obs, torch::Tensor sum,
-                        torch::Tensor sumsq) {
-  CHECK_DEV(obs); CHECK_CONT(obs);
-  TORCH_CHECK(sum.dtype() == torch::kFloat64 && sumsq.dtype() == torch::kFloat64);
-  des_launch_normalizer_observe(cfptr(obs), sum.data_ptr<double>(),
-                                sumsq.data_ptr<double>(), obs.size(0),
+void normalizer_observe(torch::Tensor obs, torch::Tensor scratch) {
+  CHECK_DEV(obs); CHECK_CONT(obs); CHECK_DEV(scratch); CHECK_CONT(scratch);
+  TORCH_CHECK(scratch.dtype() == torch::kFloat32 &&
+              scratch.numel() >= 256 * 2 * obs.size(1),
+              "observe scratch must be [256][2][dim] fp32");
+  des_launch_normalizer_observe(cfptr(obs), fptr(scratch), obs.size(0),
                                 obs.size(1), cur_stream());
+}
+
+void normalizer_observe_flush(torch::Tensor scratch, int64_t P,
+                              torch::Tensor sum, torch::Tensor sumsq) {
+  CHECK_DEV(scratch); CHECK_CONT(scratch);
+  TORCH_CHECK(sum.dtype() == torch::kFloat64 && sumsq.dtype() == torch::kFloat64);
+  int dim = sum.numel();
+  des_launch_normalizer_observe_flush(cfptr(scratch), sum.data_ptr<double>(),
+                                      sumsq.data_ptr<double>(), (int)P, dim,
+                                      cur_stream());
 }
 
 }  // namespace
@@ -375,6 +387,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("humanoid_reset", &humanoid_reset);
   m.def("humanoid_step", &humanoid_step);
   m.def("normalizer_observe", &normalizer_observe);
+  m.def("normalizer_observe_flush", &normalizer_observe_flush);
   m.def("centered_rank", &centered_rank, "K9: stable centered-rank weights");
   m.def("impala_conv", &impala_conv, "K6: 3x3 pad-1 conv (pop, tiled)");
   m.def("impala_resblock", &impala_resblock,

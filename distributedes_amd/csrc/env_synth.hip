@@ -120,7 +120,10 @@ __global__ void humanoid_step_kernel(float* __restrict__ s,
   if (i >= P) return;
   float* srow = s + (int64_t)i * HDIM;
   const float* arow = actions + (int64_t)i * HACT;
-  for (int j = lane; j < HDIM; j += 64) sold[j] = srow[j];
+  // float4 staging (HDIM = 94 * 4)
+  for (int j4 = lane; j4 < HDIM / 4; j4 += 64)
+    *reinterpret_cast<float4*>(sold + 4 * j4) =
+        *reinterpret_cast<const float4*>(srow + 4 * j4);
   if (lane < HACT) areg[lane] = arow[lane];
   __syncthreads();
 
@@ -129,15 +132,25 @@ __global__ void humanoid_step_kernel(float* __restrict__ s,
   // learning curves rise over hundreds of generations (no trivial ceiling)
   float t = (float)(steps[i] + 1);
   float rpart = 0.f, apart = 0.f;
-  for (int j = lane; j < HDIM; j += 64) {
-    float sr = sold[(j + HDIM - 1) % HDIM];
-    float u = areg[j % HACT];
-    float sn = tanhf(1.2f * sold[j] + 0.6f * sr + 0.3f * u);
-    srow[j] = sn;
-    if (j < HACT) {
-      float target = 0.6f * sinf(0.05f * t + 0.37f * (float)j);
-      float e = sn - target;
-      rpart += e * e;
+  for (int j4 = lane; j4 < HDIM / 4; j4 += 64) {
+    int j = 4 * j4;
+    float4 so = *reinterpret_cast<const float4*>(sold + j);
+    float sm1 = sold[(j + HDIM - 1) % HDIM];
+    float4 sn;
+    sn.x = tanhf(1.2f * so.x + 0.6f * sm1 + 0.3f * areg[j % HACT]);
+    sn.y = tanhf(1.2f * so.y + 0.6f * so.x + 0.3f * areg[(j + 1) % HACT]);
+    sn.z = tanhf(1.2f * so.z + 0.6f * so.y + 0.3f * areg[(j + 2) % HACT]);
+    sn.w = tanhf(1.2f * so.w + 0.6f * so.z + 0.3f * areg[(j + 3) % HACT]);
+    *reinterpret_cast<float4*>(srow + j) = sn;
+    if (j < HACT) {  // j in {0,4,8,12,16}: dims < 17 contribute tracking err
+      const float snv[4] = {sn.x, sn.y, sn.z, sn.w};
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+        if (j + q < HACT) {
+          float target = 0.6f * sinf(0.05f * t + 0.37f * (float)(j + q));
+          float e = snv[q] - target;
+          rpart += e * e;
+        }
     }
   }
   if (lane < HACT) apart = areg[lane] * areg[lane];

@@ -15,7 +15,7 @@ namespace {
 constexpr int MAX_WIDTH = 512;  // max(in_dim, hidden, A)
 
 template <typename WT, bool CONTINUOUS>
-__global__ __launch_bounds__(256) void mlp_pop_forward_kernel(
+__global__ __launch_bounds__(256, 6) void mlp_pop_forward_kernel(
     const WT* __restrict__ popw, int64_t row_stride,      // [P, >=D]
     const float* __restrict__ obs, int in_dim,            // [P, in_dim] raw
     const float* __restrict__ nmean,                      // nullable [in_dim]

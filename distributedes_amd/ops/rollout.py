@@ -153,6 +153,9 @@ class HipRollout:
         elif e == "humanoid_synth":
             self.hu_s = z(P, 376)
             self.hu_steps = z(P, dt=torch.int32)
+        if len(self.env.obs_shape) == 1:
+            # K12 two-stage observe: per-block fp32 partial buffer
+            self.nz_scratch = z(256, 2, self.env.obs_dim)
 
     # ------------------------------------------------------------------ #
     def _reset(self, gen_seed: int, env_ids: torch.Tensor):
@@ -271,12 +274,13 @@ class HipRollout:
                                    self.env.horizon)
 
     def _observe(self, normalizer):
-        """K12: fold current obs into the normalizer's fp64 delta buffers
-        (the count is added analytically in run(): graph-capture-safe)."""
+        """K12 step half: accumulate obs into the fp32 per-block partial
+        scratch; the fp64 reduce runs ONCE per generation in run() (flush) —
+        the count is added analytically there too (graph-capture-safe)."""
         if not hasattr(normalizer, "_dsum"):
             return
         obs = self.st_state if self.env_name == "cartpole" else self.hu_s
-        self.ext.normalizer_observe(obs, normalizer._dsum, normalizer._dsumsq)
+        self.ext.normalizer_observe(obs, self.nz_scratch)
 
     # ------------------------------------------------------------------ #
     def _loop(self, pop_w, episode_steps, normalizer, is_state_env,
@@ -297,6 +301,10 @@ class HipRollout:
         if self.P != P:
             self._alloc(P)
         is_state_env = len(self.env.obs_shape) == 1
+        do_observe = (is_state_env and observe
+                      and hasattr(normalizer, "_dsum"))
+        if do_observe:
+            self.nz_scratch.zero_()
         self._reset(gen_seed, env_ids)
         use_graph = (allow_graph
                      and getattr(self.cfg, "use_hip_graphs", False)
@@ -322,7 +330,10 @@ class HipRollout:
         if not use_graph:
             self._loop(pop_w, episode_steps, normalizer, is_state_env, observe)
         self.last_used_graph = use_graph
-        if is_state_env and observe and hasattr(normalizer, "_dcount"):
+        if do_observe:
+            self.ext.normalizer_observe_flush(self.nz_scratch, P,
+                                              normalizer._dsum,
+                                              normalizer._dsumsq)
             normalizer._dcount += P * episode_steps
         return self.fitness.clone(), P * episode_steps, float(self.ep_len.mean())
 
@@ -337,11 +348,11 @@ class HipRollout:
         with torch.cuda.stream(s):
             self._loop(pop_w, episode_steps, normalizer, is_state_env, observe)
         torch.cuda.current_stream().wait_stream(s)
-        # undo warmup side effects
+        # undo warmup side effects (the warmup accumulates into the observe
+        # scratch; the fp64 deltas are only touched by the post-loop flush)
         self._reset(self._gen_seed, self._ids)
         if hasattr(normalizer, "_dsum"):
-            normalizer._dsum.zero_()
-            normalizer._dsumsq.zero_()
+            self.nz_scratch.zero_()
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
             self._loop(pop_w, episode_steps, normalizer, is_state_env, observe)
