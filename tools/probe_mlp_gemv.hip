@@ -117,6 +117,72 @@ __device__ void gemv8(const unsigned short* w, const unsigned short* b,
   }
 }
 
+// K4 A/B: FUSED perturb-forward gemv — weights materialize on the fly as
+// theta[k] + s*sigma*table[off + k].  theta (651 KB fp32) is shared by all
+// members -> L2-resident regular loads; the noise slice streams nt like the
+// materialized population rows would.  HBM traffic is identical to the
+// materialized bf16 population (bf16 table), so this measures the VALU/L2
+// overhead fusion adds — and what removing the perturb kernel + 3.3 GB of
+// population residency would buy.
+template <int ACT>
+__device__ void gemv4_fused(const float* theta, const unsigned short* noise,
+                            float ssig, const float* x, float* y, int in_dim,
+                            int out_dim, int lane, int wid) {
+  for (int r0 = wid * 4; r0 < out_dim; r0 += 16) {
+    int rc = out_dim - 1;
+    float acc[4] = {0, 0, 0, 0};
+    int rows[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) rows[j] = min(r0 + j, rc);
+    for (int i = lane * 8; i < in_dim; i += 512) {
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        int64_t base = (int64_t)rows[rr] * in_dim + i;
+        u32x4 q = ld16<true>(noise + base);
+        float4 t0 = *reinterpret_cast<const float4*>(theta + base);
+        float4 t1 = *reinterpret_cast<const float4*>(theta + base + 4);
+        const unsigned short* nh = reinterpret_cast<const unsigned short*>(&q);
+        const float* tf = reinterpret_cast<const float*>(&t0);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[rr] += x[i + j] * (tf[j] + ssig * b2f(nh[j]));
+        const float* tg = reinterpret_cast<const float*>(&t1);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[rr] += x[i + 4 + j] * (tg[j] + ssig * b2f(nh[4 + j]));
+      }
+    }
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) acc[rr] = wave_sum(acc[rr]);
+    if (lane < 4 && r0 + lane < out_dim) {
+      float a = acc[lane];
+      if (ACT) a = tanhf(a);
+      y[r0 + lane] = a;
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void fwd_fused_kernel(
+    const float* __restrict__ theta, const unsigned short* __restrict__ table,
+    const long* __restrict__ offs, const float* __restrict__ signs,
+    const float* __restrict__ obs, float* __restrict__ out, int P) {
+  __shared__ float xb[512];
+  __shared__ float yb[512];
+  int i = blockIdx.x;
+  const unsigned short* noise = table + offs[i];
+  float ssig = signs[i] * 0.02f;
+  int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  for (int d = tid; d < IN; d += 256) xb[d] = obs[(int64_t)i * IN + d];
+  __syncthreads();
+  gemv4_fused<1>(theta + OW1, table + offs[i] + OW1, ssig, xb, yb, IN, H, lane, wid);
+  __syncthreads();
+  gemv4_fused<1>(theta + OW2, noise + OW2, ssig, yb, xb, H, H, lane, wid);
+  __syncthreads();
+  gemv4_fused<0>(theta + OWH, noise + OWH, ssig, xb, yb, H, A, lane, wid);
+  __syncthreads();
+  for (int a = tid; a < A; a += 256) out[(int64_t)i * A + a] = tanhf(yb[a]);
+}
+
 template <int V>
 __global__ __launch_bounds__(256) void fwd_kernel(const unsigned short* __restrict__ popw,
                                                   const float* __restrict__ obs,
@@ -173,6 +239,28 @@ int main() {
   bench("v1_gemv4", [&] { hipLaunchKernelGGL(fwd_kernel<1>, dim3(P), dim3(256), 0, 0, popw, obs, out, P); });
   bench("v2_gemv4_nt", [&] { hipLaunchKernelGGL(fwd_kernel<2>, dim3(P), dim3(256), 0, 0, popw, obs, out, P); });
   bench("v3_gemv8_nt", [&] { hipLaunchKernelGGL(fwd_kernel<3>, dim3(P), dim3(256), 0, 0, popw, obs, out, P); });
+
+  // K4 fused perturb-forward A/B (theta fp32 + bf16 noise table)
+  float* theta;
+  unsigned short* table;
+  long* offsv;
+  float* signs;
+  const int64_t TBL = 1 << 28;  // 256M bf16 = 512 MB table
+  CHK(hipMalloc(&theta, D * 4));
+  CHK(hipMalloc(&table, TBL * 2));
+  CHK(hipMalloc(&offsv, P * 8));
+  CHK(hipMalloc(&signs, P * 4));
+  CHK(hipMemset(theta, 0x3D, D * 4));
+  CHK(hipMemset(table, 0x3C, TBL * 2));
+  std::vector<long> ho(P);
+  std::vector<float> hs(P);
+  for (int i = 0; i < P; ++i) {
+    ho[i] = ((long)(i * 2654435761u) % (TBL - D - 16)) & ~15L;
+    hs[i] = (i & 1) ? -1.f : 1.f;
+  }
+  CHK(hipMemcpy(offsv, ho.data(), P * 8, hipMemcpyHostToDevice));
+  CHK(hipMemcpy(signs, hs.data(), P * 4, hipMemcpyHostToDevice));
+  bench("v4_fused_k4", [&] { hipLaunchKernelGGL(fwd_fused_kernel, dim3(P), dim3(256), 0, 0, theta, table, offsv, signs, obs, out, P); });
   CHK(hipDeviceSynchronize());
   return 0;
 }
