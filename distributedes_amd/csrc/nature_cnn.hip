@@ -41,10 +41,13 @@ constexpr int O2 = 9, S2 = 2;                     // conv2 4x4 s2, out 64ch
 constexpr int O3 = 7;                             // conv3 3x3 s1, out 64ch
 constexpr int FC_IN = 3136, FC_OUT = 512, MAX_A = 32;
 
-// LDS carve (bytes, 16-aligned): A-slab (max conv1 400x32 bf16), B-slab
-// (64x32), a1, a2; fc input/out + head overlay a1 (dead after conv2)
-constexpr int OFF_SA = 0;                         // [400][32] bf16 = 25600
-constexpr int OFF_SB = OFF_SA + 400 * 64;         // [64][32] bf16 = 4096
+// LDS carve (bytes, 16-aligned): A-slab (max conv2 96x32 bf16 — conv1 is
+// DIRECT-FRAGMENT from the L2-warm frames, no A-slab), B-slab (64x32), a1,
+// a2; fc input/out + head overlay a1 (dead after conv2).  Shrinking the
+// carve from ~66 KB to ~46 KB lifts residency to 3 blocks/CU, i.e. 24
+// waves streaming during the HBM-bound FC phase (was 16).
+constexpr int OFF_SA = 0;                         // [96][32] bf16 = 6144
+constexpr int OFF_SB = OFF_SA + 96 * 64;          // [64][32] bf16 = 4096
 constexpr int OFF_A1 = OFF_SB + 64 * 64;          // [32][400] bf16 = 25600
 constexpr int OFF_A2 = OFF_A1 + 32 * 400 * 2;     // [64][81] bf16 = 10368
 constexpr int CONV_LDS = (OFF_A2 + 64 * 81 * 2 + 15) & ~15;
@@ -99,7 +102,7 @@ __device__ __forceinline__ void tile_epilogue(const mfma_cd (&acc)[MAXT],
 }
 
 template <typename IT>
-__global__ __launch_bounds__(512, 2) void nature_conv_kernel(
+__global__ __launch_bounds__(512, 6) void nature_conv_kernel(
     const uint16_t* __restrict__ popw, int64_t row_stride,
     const IT* __restrict__ frames,     // [P][4][84][84]
     const int* __restrict__ slot_map,  // [4] logical->physical slot
@@ -134,6 +137,11 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
             s3 = slot_map[3];
 
   // ======================= conv1: 400 x 32 x 256 =========================
+  // DIRECT-FRAGMENT: each MFMA A-fragment (8 consecutive k of one K-step)
+  // is 8 contiguous pixels of one frame row — loaded straight from the
+  // L2-warm frames (28 KB/member) and packed in registers.  No A-slab, no
+  // staging pass; only the tiny B-slab stays in LDS (the freed 25.6 KB
+  // lifts whole-kernel residency to 3 blocks/CU for the FC stream).
   {
     const uint16_t* w1 = w + offs[0];
     mfma_cd acc[7];
@@ -143,37 +151,46 @@ __global__ __launch_bounds__(512, 2) void nature_conv_kernel(
       const int ic = ks >> 1, ky0 = (ks & 1) << 2;
       const int slot = (ic == 0 ? s0 : ic == 1 ? s1m : ic == 2 ? s2m : s3);
       const IT* plane = obs + slot * (HW * HW);
-      // A-slab: unit = (position m, ky row) -> 8 px -> 8 bf16 (16 B)
-      for (int e = tid; e < 400 * 4; e += 512) {
-        int m = e >> 2, kyl = e & 3;
-        int oy = m / O1, ox = m % O1;
-        const IT* row = plane + (oy * S1 + ky0 + kyl) * HW + ox * S1;
-        uint4 pk;
-        if constexpr (sizeof(IT) == 1) {
-          uint32_t d0 = *reinterpret_cast<const uint32_t*>(row);
-          uint32_t d1 = *reinterpret_cast<const uint32_t*>(row + 4);
-          pk.x = __builtin_bit_cast(uint32_t,
-                 des::pack_bf16((float)(d0 & 0xFF), (float)((d0 >> 8) & 0xFF)));
-          pk.y = __builtin_bit_cast(uint32_t,
-                 des::pack_bf16((float)((d0 >> 16) & 0xFF), (float)(d0 >> 24)));
-          pk.z = __builtin_bit_cast(uint32_t,
-                 des::pack_bf16((float)(d1 & 0xFF), (float)((d1 >> 8) & 0xFF)));
-          pk.w = __builtin_bit_cast(uint32_t,
-                 des::pack_bf16((float)((d1 >> 16) & 0xFF), (float)(d1 >> 24)));
-        } else {
-          uint2 d0 = *reinterpret_cast<const uint2*>(row);
-          uint2 d1 = *reinterpret_cast<const uint2*>(row + 4);
-          pk = {d0.x, d0.y, d1.x, d1.y};
-        }
-        *reinterpret_cast<uint4*>(sa + m * 32 + kyl * 8) = pk;
-      }
       // B-slab: [32 oc][32 k] from w1 rows (K = 256)
       for (int e = tid; e < 32 * 32; e += 512) {
         int n = e >> 5, kk = e & 31;
         sb[n * 32 + kk] = w1[n * 256 + ks * 32 + kk];
       }
       __syncthreads();
-      mfma_tiles<7>(sa, sb, acc, wid, lane, 50, 2);
+      const int kyl = lane >> 4;       // fragment chunk = one ky row
+#pragma unroll
+      for (int u = 0; u < 7; ++u) {
+        int t = wid + 8 * u;
+        if (t < 50) {
+          int mt = t / 2, nt = t % 2;
+          int m = mt * 16 + (lane & 15);
+          int oy = m / O1, ox = m % O1;
+          const IT* row = plane + (oy * S1 + ky0 + kyl) * HW + ox * S1;
+          mfma_ab af;
+          if constexpr (sizeof(IT) == 1) {
+            uint32_t d0 = *reinterpret_cast<const uint32_t*>(row);
+            uint32_t d1 = *reinterpret_cast<const uint32_t*>(row + 4);
+            uint4 pk;
+            pk.x = __builtin_bit_cast(uint32_t,
+                   des::pack_bf16((float)(d0 & 0xFF), (float)((d0 >> 8) & 0xFF)));
+            pk.y = __builtin_bit_cast(uint32_t,
+                   des::pack_bf16((float)((d0 >> 16) & 0xFF), (float)(d0 >> 24)));
+            pk.z = __builtin_bit_cast(uint32_t,
+                   des::pack_bf16((float)(d1 & 0xFF), (float)((d1 >> 8) & 0xFF)));
+            pk.w = __builtin_bit_cast(uint32_t,
+                   des::pack_bf16((float)((d1 >> 16) & 0xFF), (float)(d1 >> 24)));
+            af = __builtin_bit_cast(mfma_ab, pk);
+          } else {
+            uint2 d0 = *reinterpret_cast<const uint2*>(row);
+            uint2 d1 = *reinterpret_cast<const uint2*>(row + 4);
+            af = __builtin_bit_cast(mfma_ab, uint4{d0.x, d0.y, d1.x, d1.y});
+          }
+          mfma_ab bf = *reinterpret_cast<const mfma_ab*>(
+              sb + (nt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
+          acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[u],
+                                                           0, 0, 0);
+        }
+      }
       __syncthreads();
     }
     tile_epilogue<7>(acc, w + offs[1], a1, 400, wid, lane, 50, 2);
