@@ -42,9 +42,31 @@ def main():
     ap.add_argument("--profile", default=None, metavar="DIR",
                     help="write a torch.profiler chrome trace of the timed "
                          "steps into DIR (SURVEY.md §5.1 tracing hook)")
+    ap.add_argument("--rocprof", default=None, metavar="DIR",
+                    help="re-exec this bench under rocprofv3 --kernel-trace "
+                         "--stats with outputs in DIR (SURVEY.md §5.1b: "
+                         "rocprof counters as a harness mode, not an "
+                         "external workflow)")
     ap.add_argument("--set", nargs="*", default=[], metavar="K=V",
                     help="extra Config overrides, e.g. --set noise_dtype='bf16'")
     args = ap.parse_args()
+
+    if args.rocprof and "DES_ROCPROF_ACTIVE" not in os.environ:
+        # wrap ourselves in rocprofv3 (kernel trace + stats only: counter
+        # collection must never be combined with trace domains, see repo
+        # profiling notes); cwd moves to /tmp because rocprofv3 scratches
+        # in its working directory
+        import subprocess
+        import sys
+        outdir = os.path.abspath(args.rocprof)
+        os.makedirs(outdir, exist_ok=True)
+        script = os.path.abspath(__file__)
+        argv = [a for a in sys.argv[1:]
+                if not (a == "--rocprof" or a == args.rocprof)]
+        env = dict(os.environ, DES_ROCPROF_ACTIVE="1", TMPDIR="/tmp")
+        cmd = ["rocprofv3", "--kernel-trace", "--stats", "-d", outdir,
+               "-o", "bench", "--", sys.executable, script] + argv
+        raise SystemExit(subprocess.call(cmd, env=env, cwd="/tmp"))
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))

@@ -177,3 +177,18 @@ def test_render_filmstrip(tmp_path):
     assert len(frames) == 3 and frames[0].shape == (84, 84)
     svg = frames_to_svg(frames, rewards)
     assert "<svg" in svg
+
+
+def test_atari57_per_game_fitness_reported():
+    """BASELINE config #4: the engine reports per-game fitness for the
+    57-game multi-task population (engine.train_step)."""
+    from distributedes_amd import preset
+    from distributedes_amd.core.engine import TrainState, train_step
+    cfg = preset("atari57_es", pop=8, episode_steps=2, device="cpu",
+                 noise_table_size=1 << 21, log_every=10 ** 9)
+    st = TrainState(cfg, ops_backend="torch_ref")
+    s = train_step(st)
+    pg = s["per_game_return_mean"]
+    assert len(pg) == 57
+    # pop=8 antithetic -> env ids 0..3 -> games 0..3 populated
+    assert all(pg[g] == 0.0 for g in range(4, 57))
