@@ -68,3 +68,52 @@ def test_rollout_fitness_finite_and_varied():
     fitness, env_steps, _ = st.rollout(pop_w)
     assert torch.isfinite(fitness).all()
     assert env_steps == 64 * 60
+
+
+def test_hip_graph_replay_matches_eager_across_generations():
+    """Round-1 advisor HIGH finding: captured graphs baked a freeable env_id
+    pointer, corrupting per-env RNG streams from the 2nd generation after
+    capture.  Two identical runs — graphs on vs off — must produce the
+    SAME fitness sequence over several generations (the corruption only
+    shows up on replays after the first)."""
+    def run(graphs):
+        cfg = preset("pong_es_8k", pop=64, episode_steps=10,
+                     noise_table_size=1 << 22, device="cuda",
+                     use_hip_graphs=graphs, log_every=10 ** 9)
+        st = TrainState(cfg)
+        rets = []
+        for _ in range(4):
+            rets.append(train_step(st)["episodic_return_mean"])
+        if graphs:
+            assert st._hip_rollout.last_used_graph, "graphs must engage"
+        return rets, st.theta.cpu()
+
+    r_g, th_g = run(True)
+    r_e, th_e = run(False)
+    assert r_g == r_e, (r_g, r_e)
+    torch.testing.assert_close(th_g, th_e, rtol=0, atol=0)
+
+
+def test_evaluate_does_not_touch_training_state():
+    """Round-1 advisor MEDIUM finding: evaluate() must not fold eval
+    observations into the normalizer deltas nor mutate TrainState."""
+    from distributedes_amd import evaluate
+    cfg = preset("cartpole_es", pop=16, episode_steps=10,
+                 noise_table_size=1 << 20, device="cuda", log_every=10 ** 9)
+    st = TrainState(cfg)
+    train_step(st)
+    before = {
+        "dsum": st.normalizer._dsum.clone(),
+        "dcount": st.normalizer._dcount.clone(),
+        "member_base": st.member_base,
+        "local_pop": st.local_pop,
+        "cfg": st.cfg,
+        "theta": st.theta.clone(),
+    }
+    evaluate(st, episodes=8)
+    assert torch.equal(before["dsum"], st.normalizer._dsum)
+    assert torch.equal(before["dcount"], st.normalizer._dcount)
+    assert st.member_base == before["member_base"]
+    assert st.local_pop == before["local_pop"]
+    assert st.cfg is before["cfg"]
+    assert torch.equal(before["theta"], st.theta)
