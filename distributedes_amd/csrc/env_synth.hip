@@ -36,6 +36,9 @@ template <int C, int HW>
 __global__ void latent_render_kernel(const float* __restrict__ z,
                                      const float* __restrict__ coef,
                                      uint16_t* __restrict__ obs, int P) {
+  // 8 contiguous pixels per unit -> uint4 stores (the per-pixel scalar
+  // store version ran ~5x over the write floor); HW 84/64 are 4-multiples
+  // so units never straddle rows beyond the handled tail
   int i = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   int lane = threadIdx.x & 63;
   if (i >= P) return;
@@ -44,13 +47,26 @@ __global__ void latent_render_kernel(const float* __restrict__ z,
   for (int j = 0; j < ZDIM; ++j) zl[j] = z[i * ZDIM + j];
   float f0 = coef[i * 6 + 5];
   uint16_t* out = obs + (int64_t)i * C * HW * HW;
-  for (int pix = lane; pix < C * HW * HW; pix += 64) {
-    int c = pix / (HW * HW), r = pix % (HW * HW);
-    int h = r / HW, w = r % HW;
+  constexpr int NPIX = C * HW * HW;
+  for (int p8 = lane; p8 < NPIX / 8; p8 += 64) {
+    int pix0 = p8 * 8;
+    int c = pix0 / (HW * HW), r = pix0 % (HW * HW);
+    int h = r / HW, w0 = r % HW;
     float zc0 = zl[c % ZDIM], zc1 = zl[(c + 1) % ZDIM], zc2 = zl[(c + 2) % ZDIM];
-    float v = 0.5f + 0.5f * __sinf(f0 * zc0 + 0.07f * h * (1.0f + zc1) +
-                                   0.05f * w * (1.0f + zc2));
-    out[pix] = des::f32_to_bf16(v);
+    float base = f0 * zc0 + 0.07f * h * (1.0f + zc1);
+    float dx = 0.05f * (1.0f + zc2);
+    uint16_t o[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int w = w0 + j;
+      float ph = base, xw = (float)w;
+      if (w >= HW) {  // unit crosses into the next row (HW % 8 != 0)
+        xw = (float)(w - HW);
+        ph = f0 * zc0 + 0.07f * (h + 1) * (1.0f + zc1);
+      }
+      o[j] = des::f32_to_bf16(0.5f + 0.5f * __sinf(ph + dx * xw));
+    }
+    *reinterpret_cast<uint4*>(out + pix0) = *reinterpret_cast<const uint4*>(o);
   }
 }
 

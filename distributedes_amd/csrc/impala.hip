@@ -414,10 +414,17 @@ __global__ __launch_bounds__(256) void impala_fc_head_kernel(
     for (int it = 0; it < 4; ++it) {       // 2048 = 4 * 512 exactly
       int ib = lane * 8 + it * 512;
       uint4 xp = *reinterpret_cast<const uint4*>(xs + ib);
-      uint4 wa = *reinterpret_cast<const uint4*>(w0 + ib);
-      uint4 wb = *reinterpret_cast<const uint4*>(w0 + IFC_IN + ib);
-      uint4 wc = *reinterpret_cast<const uint4*>(w0 + 2 * IFC_IN + ib);
-      uint4 wd = *reinterpret_cast<const uint4*>(w0 + 3 * IFC_IN + ib);
+      // nt: fc weights are read exactly once per launch (same rationale
+      // as the Nature FC / MLP streams)
+      typedef __attribute__((ext_vector_type(4))) unsigned int ntv;
+      uint4 wa = __builtin_bit_cast(uint4, __builtin_nontemporal_load(
+          reinterpret_cast<const ntv*>(w0 + ib)));
+      uint4 wb = __builtin_bit_cast(uint4, __builtin_nontemporal_load(
+          reinterpret_cast<const ntv*>(w0 + IFC_IN + ib)));
+      uint4 wc = __builtin_bit_cast(uint4, __builtin_nontemporal_load(
+          reinterpret_cast<const ntv*>(w0 + 2 * IFC_IN + ib)));
+      uint4 wd = __builtin_bit_cast(uint4, __builtin_nontemporal_load(
+          reinterpret_cast<const ntv*>(w0 + 3 * IFC_IN + ib)));
       const uint32_t* xq = reinterpret_cast<const uint32_t*>(&xp);
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
@@ -486,8 +493,8 @@ void des_launch_impala_conv3(const uint16_t* in, uint16_t* out,
           b_off, P);                                                          \
     return;                                                                   \
   }
-  CASE(0, 3, 16, 64, 16, 32)
-  CASE(1, 16, 16, 32, 16, 32)
+  CASE(0, 3, 16, 64, 32, 32)
+  CASE(1, 16, 16, 32, 32, 32)
   CASE(2, 16, 32, 32, 16, 32)
   CASE(3, 32, 32, 16, 16, 16)
   CASE(4, 32, 32, 8, 8, 8)
