@@ -41,14 +41,14 @@ constexpr int O2 = 9, S2 = 2;                     // conv2 4x4 s2, out 64ch
 constexpr int O3 = 7;                             // conv3 3x3 s1, out 64ch
 constexpr int FC_IN = 3136, FC_OUT = 512, MAX_A = 32;
 
-// LDS carve (bytes, 16-aligned): A-slab (max conv2 96x32 bf16 — conv1 is
-// DIRECT-FRAGMENT from the L2-warm frames, no A-slab), B-slab (64x32), a1,
-// a2; fc input/out + head overlay a1 (dead after conv2).  Shrinking the
-// carve from ~66 KB to ~46 KB lifts residency to 3 blocks/CU, i.e. 24
-// waves streaming during the HBM-bound FC phase (was 16).
-constexpr int OFF_SA = 0;                         // [96][32] bf16 = 6144
-constexpr int OFF_SB = OFF_SA + 96 * 64;          // [64][32] bf16 = 4096
-constexpr int OFF_A1 = OFF_SB + 64 * 64;          // [32][400] bf16 = 25600
+// LDS carve (bytes, 16-aligned): ONLY the activations live in LDS.
+// conv1 is DIRECT-FRAGMENT from the L2-warm frames; every conv's
+// B-fragments read DIRECTLY from the member's weights (8 consecutive k of
+// an oc row = one aligned 16-B global load, L1-cached across the m-tiles
+// that share it) — no B-slab; conv2's A-slab overlays the not-yet-written
+// a2 region and conv3's the dead a1 region.  36 KB/block -> 4 blocks/CU,
+// i.e. 32 waves streaming during the HBM-bound FC phase (round 1: 16).
+constexpr int OFF_A1 = 0;                         // [32][400] bf16 = 25600
 constexpr int OFF_A2 = OFF_A1 + 32 * 400 * 2;     // [64][81] bf16 = 10368
 constexpr int CONV_LDS = (OFF_A2 + 64 * 81 * 2 + 15) & ~15;
 constexpr int OFF_A3 = OFF_A1;                    // bf16 fc input [3136]
@@ -71,6 +71,29 @@ __device__ __forceinline__ void mfma_tiles(const uint16_t* sa,
           sa + (mt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
       mfma_ab bf = *reinterpret_cast<const mfma_ab*>(
           sb + (nt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
+      acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[u], 0, 0, 0);
+    }
+  }
+}
+
+// mfma step with the B-fragment read directly from the member's global
+// weight rows (ldb = K of the layer); the same fragment is L1-hot for the
+// m-tiles sharing an nt
+template <int MAXT>
+__device__ __forceinline__ void mfma_tiles_gb(const uint16_t* sa,
+                                              const uint16_t* gb, int ldb,
+                                              int k0, mfma_cd (&acc)[MAXT],
+                                              int wid, int lane, int ntiles,
+                                              int ncols) {
+#pragma unroll
+  for (int u = 0; u < MAXT; ++u) {
+    int t = wid + 8 * u;
+    if (t < ntiles) {
+      int mt = t / ncols, nt = t % ncols;
+      mfma_ab af = *reinterpret_cast<const mfma_ab*>(
+          sa + (mt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
+      mfma_ab bf = *reinterpret_cast<const mfma_ab*>(
+          gb + (int64_t)(nt * 16 + (lane & 15)) * ldb + k0 + (lane >> 4) * 8);
       acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[u], 0, 0, 0);
     }
   }
@@ -101,18 +124,18 @@ __device__ __forceinline__ void tile_epilogue(const mfma_cd (&acc)[MAXT],
   }
 }
 
-template <typename IT>
-__global__ __launch_bounds__(512, 6) void nature_conv_kernel(
+template <typename IT, int MINW = 6>
+__global__ __launch_bounds__(512, MINW) void nature_conv_kernel(
     const uint16_t* __restrict__ popw, int64_t row_stride,
     const IT* __restrict__ frames,     // [P][4][84][84]
     const int* __restrict__ slot_map,  // [4] logical->physical slot
     int* __restrict__ actions, float* __restrict__ logits_out, int A, int P,
     const int64_t* __restrict__ offs, unsigned stagger) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  uint16_t* sa = reinterpret_cast<uint16_t*>(smem + OFF_SA);
-  uint16_t* sb = reinterpret_cast<uint16_t*>(smem + OFF_SB);
   uint16_t* a1 = reinterpret_cast<uint16_t*>(smem + OFF_A1);
   uint16_t* a2 = reinterpret_cast<uint16_t*>(smem + OFF_A2);
+  uint16_t* sa2 = a2;   // conv2 A-slab: a2 not written until conv2 epilogue
+  uint16_t* sa3 = a1;   // conv3 A-slab: a1 dead once conv2 is done
   uint16_t* a3 = reinterpret_cast<uint16_t*>(smem + OFF_A3);
   float* fcb = reinterpret_cast<float*>(smem + OFF_FC);
   float* hd = reinterpret_cast<float*>(smem + OFF_HD);
@@ -151,12 +174,6 @@ __global__ __launch_bounds__(512, 6) void nature_conv_kernel(
       const int ic = ks >> 1, ky0 = (ks & 1) << 2;
       const int slot = (ic == 0 ? s0 : ic == 1 ? s1m : ic == 2 ? s2m : s3);
       const IT* plane = obs + slot * (HW * HW);
-      // B-slab: [32 oc][32 k] from w1 rows (K = 256)
-      for (int e = tid; e < 32 * 32; e += 512) {
-        int n = e >> 5, kk = e & 31;
-        sb[n * 32 + kk] = w1[n * 256 + ks * 32 + kk];
-      }
-      __syncthreads();
       const int kyl = lane >> 4;       // fragment chunk = one ky row
 #pragma unroll
       for (int u = 0; u < 7; ++u) {
@@ -186,12 +203,11 @@ __global__ __launch_bounds__(512, 6) void nature_conv_kernel(
             af = __builtin_bit_cast(mfma_ab, uint4{d0.x, d0.y, d1.x, d1.y});
           }
           mfma_ab bf = *reinterpret_cast<const mfma_ab*>(
-              sb + (nt * 16 + (lane & 15)) * 32 + (lane >> 4) * 8);
+              w1 + (nt * 16 + (lane & 15)) * 256 + ks * 32 + (lane >> 4) * 8);
           acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[u],
                                                            0, 0, 0);
         }
       }
-      __syncthreads();
     }
     tile_epilogue<7>(acc, w + offs[1], a1, 400, wid, lane, 50, 2);
   }
@@ -215,15 +231,10 @@ __global__ __launch_bounds__(512, 6) void nature_conv_kernel(
           v.x = *reinterpret_cast<const uint32_t*>(ap);       // even index
           v.y = *reinterpret_cast<const uint32_t*>(ap + 2);
         }
-        *reinterpret_cast<uint2*>(sa + m * 32 + icl * 16 + ky * 4) = v;
-      }
-      // B-slab: [64 oc][32 k] (K = 512)
-      for (int e = tid; e < 64 * 32; e += 512) {
-        int n = e >> 5, kk = e & 31;
-        sb[n * 32 + kk] = w2[n * 512 + ks * 32 + kk];
+        *reinterpret_cast<uint2*>(sa2 + m * 32 + icl * 16 + ky * 4) = v;
       }
       __syncthreads();
-      mfma_tiles<3>(sa, sb, acc, wid, lane, 24, 4);
+      mfma_tiles_gb<3>(sa2, w2, 512, ks * 32, acc, wid, lane, 24, 4);
       __syncthreads();
     }
     tile_epilogue<3>(acc, w + offs[3], a2, 81, wid, lane, 24, 4);
@@ -246,11 +257,10 @@ __global__ __launch_bounds__(512, 6) void nature_conv_kernel(
           int oy = m / O3, ox = m % O3;
           v = a2[ic * 81 + (oy + ky) * O2 + ox + kx];
         }
-        sa[m * 32 + kk] = v;
-        sb[m * 32 + kk] = w3[m * 576 + icc];  // same e-split covers 64 oc
+        sa3[m * 32 + kk] = v;
       }
       __syncthreads();
-      mfma_tiles<2>(sa, sb, acc, wid, lane, 16, 4);
+      mfma_tiles_gb<2>(sa3, w3, 576, ks * 32, acc, wid, lane, 16, 4);
       __syncthreads();
     }
     // a2 fully consumed above; a3 overlays the (dead) a1 region
@@ -330,21 +340,24 @@ extern "C" void des_launch_nature_cnn_forward(
     const char* e = getenv("DES_NATURE_STAGGER");
     return e ? (unsigned)atoi(e) : 0u;
   }();
+  // 6 waves/SIMD (3 blocks/CU, zero spills) A/B-measured ahead of the
+  // 8-wave 4-block variant (1.762M vs 1.664M env steps/s: its 8 spilled
+  // VGPRs cost more than the extra streaming waves buy)
   static bool attr_set = false;
   if (!attr_set) {
-    hipFuncSetAttribute((const void*)nature_conv_kernel<uint8_t>,
+    hipFuncSetAttribute((const void*)nature_conv_kernel<uint8_t, 6>,
                         hipFuncAttributeMaxDynamicSharedMemorySize, CONV_LDS);
-    hipFuncSetAttribute((const void*)nature_conv_kernel<uint16_t>,
+    hipFuncSetAttribute((const void*)nature_conv_kernel<uint16_t, 6>,
                         hipFuncAttributeMaxDynamicSharedMemorySize, CONV_LDS);
     attr_set = true;
   }
   if (obs_u8)
-    hipLaunchKernelGGL(nature_conv_kernel<uint8_t>, dim3((uint32_t)P),
+    hipLaunchKernelGGL((nature_conv_kernel<uint8_t, 6>), dim3((uint32_t)P),
                        dim3(512), CONV_LDS, stream, popw, row_stride,
                        (const uint8_t*)frames, slot_map_dev, actions,
                        logits_out, A, P, offs_dev, stagger);
   else
-    hipLaunchKernelGGL(nature_conv_kernel<uint16_t>, dim3((uint32_t)P),
+    hipLaunchKernelGGL((nature_conv_kernel<uint16_t, 6>), dim3((uint32_t)P),
                        dim3(512), CONV_LDS, stream, popw, row_stride,
                        (const uint16_t*)frames, slot_map_dev, actions,
                        logits_out, A, P, offs_dev, stagger);
