@@ -30,6 +30,13 @@ __device__ __forceinline__ float4 load4(const float* p) {     // fp32x4
 __device__ __forceinline__ float load1(const uint16_t* p) { return bf16_to_f32(*p); }
 __device__ __forceinline__ float load1(const float* p) { return *p; }
 
+// NOTE: do NOT add __builtin_nontemporal_load(ext_vector*) helpers that
+// scalarize their result: on this toolchain that pattern can lower to a
+// broadcast of element 0 (reproduced by tools/probe_nt_load.hip) and
+// silently corrupted the perturb/wsum table streams when tried.  The nt
+// loads in gemv_rows4_w8 / the CNN FC streams consume the vector whole
+// and are parity-tested (incl. a bit-exact golden pin).
+
 // y[row] over rows [wid, wid+NW, ...): act(dot(W[row], x) + b[row])
 // W rows at w + row*stride (elements), x fp32 in LDS, y fp32 in LDS.
 // ACT: 0 = identity, 1 = tanh, 2 = relu
