@@ -111,3 +111,29 @@ def test_ws2_ranks_bitwise_identical_and_match_ws1():
 def test_ws2_resume_bit_equivalence():
     res = _spawn("_body_resume", world=2)
     assert res[0]["same"] and res[1]["same"]
+
+
+def _body_atari57(rank, world):
+    from distributedes_amd import preset as mkpreset
+    from distributedes_amd.core.engine import TrainState, train_step
+    cfg = mkpreset("atari57_es", pop=8, episode_steps=2, device="cpu",
+                   backend="gloo", noise_table_size=1 << 21,
+                   log_every=10 ** 9, metrics_path=None, checkpoint_every=0)
+    st = TrainState(cfg, ops_backend="torch_ref")
+    s = train_step(st)
+    return {"per_game": s["per_game_return_mean"],
+            "theta": st.theta.clone()}
+
+
+def test_atari57_ws2_per_game_stats_rank_identical():
+    """BASELINE config #4 on the distributed tier: the 57-game multi-task
+    per-game fitness breakdown is computed from the all-gathered GLOBAL
+    fitness, so every rank must report the identical 57-vector, and theta
+    must stay bitwise rank-identical."""
+    res = _spawn("_body_atari57", world=2)
+    assert res[0]["per_game"] == res[1]["per_game"]
+    assert len(res[0]["per_game"]) == 57
+    assert torch.equal(res[0]["theta"], res[1]["theta"])
+    # pop=8 antithetic -> 4 env streams -> exactly games 0..3 are populated
+    assert any(v != 0.0 for v in res[0]["per_game"][:4])
+    assert all(v == 0.0 for v in res[0]["per_game"][4:])
