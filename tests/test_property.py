@@ -54,3 +54,36 @@ def test_member_offsets_always_in_bounds(seed, gen, pop, anti):
     assert (offs >= 0).all() and (offs + D <= T).all()
     assert (offs % 16 == 0).all()
     assert set(np.unique(signs)) <= {-1.0, 1.0}
+
+
+@given(st.integers(0, 2**30), st.integers(2, 57))
+@settings(max_examples=30, deadline=None)
+def test_game_coefs_deterministic_and_in_range(seed, ng):
+    """Atari-57 parameterization invariants: hashed per-game coefficients
+    are pure functions of the game id, stay in their documented ranges,
+    and distinct games get distinct rows (task distinctness)."""
+    import torch
+
+    from distributedes_amd.envs.synthetic import game_coefs
+    g = torch.arange(ng, dtype=torch.int64)
+    a = game_coefs(g, ng)
+    b = game_coefs(g.clone(), ng)
+    assert torch.equal(a, b)
+    lo = torch.tensor([1.10, 0.40, 0.05, 1.50, 0.20, 2.00])
+    hi = torch.tensor([1.60, 0.90, 0.30, 4.50, 0.80, 4.00])
+    assert (a >= lo - 1e-6).all() and (a <= hi + 1e-6).all()
+    assert torch.unique(a, dim=0).shape[0] == ng
+
+
+@given(st.integers(0, 2**20), st.integers(1, 40))
+@settings(max_examples=30, deadline=None)
+def test_antithetic_pair_shares_offset_with_opposite_sign(seed, gen):
+    """Mirrored sampling invariant: members 2k and 2k+1 draw the SAME
+    noise slice with opposite signs (the variance-reduction contract the
+    ES update and the env-stream pairing both rely on)."""
+    import numpy as np
+
+    from distributedes_amd.core.noise import member_offsets
+    offs, signs = member_offsets(seed, gen, 16, True, 1 << 20, 1024)
+    assert np.array_equal(offs[0::2], offs[1::2])
+    assert np.all(signs[0::2] == 1.0) and np.all(signs[1::2] == -1.0)
